@@ -1,0 +1,141 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: distributed SigLIP contrastive training step.
+
+Measures the BASELINE.json north-star metric — image-text pairs/sec,
+forward+backward(+grad-average+optimizer), global batch B=32768, emb dim 768,
+bf16, strong scaling over 1/2/4/8 MI355X GPUs (per-rank batch = B/N; the
+pair-grid work B² is fixed as N grows).
+
+One step = encode both towers → L2-normalize → DistributedSigmoidLoss
+(fused CDNA4 HIP kernels; ring or all-gather comm over RCCL/xGMI) → backward →
+manual DDP gradient averaging → SGD step.  Data is synthetic random features;
+weights are random-init (no datasets/checkpoints are available offline).
+
+Launch (driver contract):
+    python bench.py --gpus N --steps K --warmup W
+    # N>1: python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+    #        --master-addr 127.0.0.1 bench.py --gpus N ...
+Rank 0 prints exactly one JSON line with the aggregate metric.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import time
+
+import torch
+import torch.distributed as dist
+
+from distributed_sigmoid_loss_amd import DistributedSigmoidLoss
+from distributed_sigmoid_loss_amd.models import TwoTowerModel
+from distributed_sigmoid_loss_amd.parallel import average_gradients
+from distributed_sigmoid_loss_amd.utils import init_from_env, set_seed
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1,
+                   help="world size (informational; actual size from env)")
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--global-batch", type=int, default=32768)
+    p.add_argument("--dim", type=int, default=768)
+    p.add_argument("--strategy", choices=["ring", "all_gather"],
+                   default="ring")
+    p.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
+    p.add_argument("--device", choices=["cuda", "cpu"], default=None)
+    p.add_argument("--col-chunk", type=int, default=None,
+                   help="column slab size for chunked negatives")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank, local_rank, world = init_from_env()
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    if device == "cpu" and args.dtype == "bf16":
+        args.dtype = "fp32"  # CPU plumbing config runs fp32
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+
+    if args.global_batch % world:
+        raise SystemExit(f"global batch {args.global_batch} not divisible by "
+                         f"world size {world}")
+    b = args.global_batch // world
+    set_seed(1234 + rank)
+
+    model = TwoTowerModel(args.dim, args.dim).to(device=device, dtype=dtype)
+    loss_mod = DistributedSigmoidLoss(b, strategy=args.strategy,
+                                      col_chunk=args.col_chunk).to(device)
+    params = list(model.parameters()) + list(loss_mod.parameters())
+    opt = torch.optim.SGD(params, lr=1e-4)
+
+    img_feats = torch.randn(b, args.dim, device=device, dtype=dtype)
+    txt_feats = torch.randn(b, args.dim, device=device, dtype=dtype)
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        zi, zt = model(img_feats, txt_feats)
+        loss = loss_mod(zi, zt)
+        loss.backward()
+        if world > 1:
+            average_gradients(model)
+            average_gradients(loss_mod)
+        opt.step()
+        return loss
+
+    def sync():
+        if device == "cuda":
+            torch.cuda.synchronize()
+        if world > 1:
+            dist.barrier()
+
+    for _ in range(args.warmup):
+        step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # Max over ranks so stragglers count.
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if dist.get_backend() != "gloo":
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    pairs_per_sec = args.global_batch / (elapsed / args.steps)
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "image-text pairs/sec fwd+bwd",
+            "value": pairs_per_sec,
+            "unit": "pairs/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": "two-tower-linear-siglip",
+                "global_batch": args.global_batch,
+                "emb_dim": args.dim,
+                "seq_len": None,
+                "parallelism": f"dp{world}-{args.strategy}",
+            },
+        }), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,277 @@
+"""Distributed SigLIP loss modules.
+
+Two public modules, mirroring the reference repo's API surface:
+
+- :class:`DistributedSigmoidLoss` — module-owned learnable ``t_prime`` (init
+  ``log 10``) and ``bias`` (init ``−10``); forward takes the *local* image and
+  text embedding shards and returns ``Σ loss / gpu_batch_size``.  Behavioral
+  parity with reference ``distributed_sigmoid_loss.py:8-48``.  Strategy is
+  selectable: ``"all_gather"`` (differentiable all-gather, the reference's
+  scheme) or ``"ring"`` (pipelined P2P ring with comm/compute overlap — the
+  MI355X performance path).
+- :class:`SigLipLoss` — caller-owned ``logit_scale``/``logit_bias`` passed to
+  ``forward`` as arguments; ring strategy via the autograd neighbour-exchange
+  primitives.  Behavioral parity with reference
+  ``rwightman_sigmoid_loss.py:12-124`` (including ``bidir`` and the
+  ``output_dict`` return form).
+
+Unlike the reference (which builds labels on the default CPU device,
+``distributed_sigmoid_loss.py:28-30``), everything here is device- and
+dtype-correct: labels are an ``i == j + offset`` index predicate, never a
+materialized tensor on the hot path.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.distributed as dist
+
+from .functional import (
+    sigmoid_contrastive_loss,
+    chunk_loss_fwd,
+    chunk_loss_bwd,
+    DEFAULT_COL_CHUNK,
+)
+from ..parallel.collectives import all_gather_with_grad, _backend_is_gloo
+from ..parallel.ring import (
+    neighbour_exchange_with_grad,
+    neighbour_exchange_bidir_with_grad,
+    neighbour_exchange_start,
+)
+
+
+def _world_and_rank(group=None):
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_world_size(group), dist.get_rank(group)
+    return 1, 0
+
+
+class _RingAllGatherLoss(torch.autograd.Function):
+    """Fused distributed loss: ring-pipelined forward, reduce-scatter backward.
+
+    Forward walks the text shards around a unidirectional P2P ring; the
+    exchange of chunk k+1 is posted *before* the loss kernel for chunk k runs,
+    so on RCCL the xGMI wire time hides under the MFMA compute (the reference
+    serializes these — ``distributed_utils.py:25-26``).  All received shards
+    are kept (O(W·b·d), trivial against 288 GB HBM3E) so backward never
+    re-communicates embeddings.
+
+    Backward computes, per source shard s, the local gradient contribution
+    dztxt_s (recomputing logit tiles via the fused HIP kernel), concatenates
+    them in rank order and issues ONE ``reduce_scatter_tensor(SUM)`` — the
+    algebraic collapse of the reference's W−1 reversed autograd ring hops
+    (``distributed_utils.py:74-77``) into a single RCCL collective.
+
+    Gradient semantics match the differentiable all-gather strategy exactly:
+    raw per-rank grads already carry full cross-rank contributions (what the
+    reference's strategy-equivalence oracle asserts,
+    ``test_sigmoid_loss_variants.py:112-113``).
+    """
+
+    @staticmethod
+    def forward(ctx, zimg, ztxt, t_prime, bias, group, col_chunk, impl):
+        world, rank = _world_and_rank(group)
+        zimg = zimg.contiguous()
+        ztxt = ztxt.contiguous()
+        b_txt = ztxt.shape[0]
+
+        chunks = [None] * world          # indexed by source rank
+        chunks[rank] = ztxt
+
+        if world > 1:
+            left = (rank - 1 + world) % world
+            right = (rank + 1) % world
+            # Post hop 1 before any compute: its wire time hides under the
+            # local-block kernel below.
+            handle = neighbour_exchange_start(left, right, ztxt, group=group)
+
+        loss = chunk_loss_fwd(zimg, ztxt, t_prime, bias,
+                              diag_offset=0, col_chunk=col_chunk, impl=impl)
+
+        if world > 1:
+            for hop in range(1, world):
+                # Post hop k+1 before computing on hop k's data.
+                recv = handle.wait()[0]
+                src = (rank - hop + world) % world
+                chunks[src] = recv
+                if hop < world - 1:
+                    handle = neighbour_exchange_start(left, right, recv,
+                                                      group=group)
+                loss = loss + chunk_loss_fwd(zimg, recv, t_prime, bias,
+                                             diag_offset=None,
+                                             col_chunk=col_chunk, impl=impl)
+
+        ctx.save_for_backward(zimg, t_prime, bias, *chunks)
+        ctx.group = group
+        ctx.world = world
+        ctx.rank = rank
+        ctx.col_chunk = col_chunk
+        ctx.impl = impl
+        ctx.b_txt = b_txt
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        zimg, t_prime, bias = ctx.saved_tensors[:3]
+        chunks = ctx.saved_tensors[3:]
+        world, rank = ctx.world, ctx.rank
+
+        dzimg = None
+        dt_prime = None
+        dbias = None
+        contribs = []
+        for s in range(world):
+            off = 0 if s == rank else None
+            dzi, dzt, dtp, dbi = chunk_loss_bwd(
+                zimg, chunks[s], t_prime, bias, off, grad_output,
+                col_chunk=ctx.col_chunk, impl=ctx.impl)
+            dzimg = dzi if dzimg is None else dzimg + dzi
+            dt_prime = dtp if dt_prime is None else dt_prime + dtp
+            dbias = dbi if dbias is None else dbias + dbi
+            contribs.append(dzt)
+
+        if world > 1:
+            flat = torch.cat(contribs, dim=0)
+            b = ctx.b_txt
+            if _backend_is_gloo(ctx.group):
+                dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=ctx.group)
+                dztxt = flat[rank * b:(rank + 1) * b].clone()
+            else:
+                dztxt = torch.empty_like(contribs[rank])
+                dist.reduce_scatter_tensor(dztxt, flat, op=dist.ReduceOp.SUM,
+                                           group=ctx.group)
+        else:
+            dztxt = contribs[0]
+
+        return dzimg, dztxt, dt_prime, dbias, None, None, None
+
+
+class DistributedSigmoidLoss(nn.Module):
+    """All-rank distributed sigmoid contrastive loss with module-owned params.
+
+    Parity with reference ``DDPSigmoidLoss`` (``distributed_sigmoid_loss.py``):
+    ``t_prime`` init ``log(10)``, ``bias`` init ``−10.0`` (``:11-12``); forward
+    all-gathers the text shard, computes ``−logsigmoid(l·z)`` over the local
+    ``(b, W·b)`` pair grid (labels ``+1`` on the rank-diagonal block's
+    diagonal, ``−1`` elsewhere) and divides by the *local* ``gpu_batch_size``
+    (``:47``).  Under DDP grad averaging this reproduces single-rank
+    ``sum/(W·b)`` normalization exactly.
+
+    ``strategy``:
+        - ``"all_gather"`` — differentiable all-gather (one RCCL all-gather
+          fwd, one reduce-scatter bwd), then ONE fused kernel call over the
+          whole ``(b, W·b)`` block.
+        - ``"ring"`` — pipelined P2P ring (xGMI point-to-point) with
+          comm/compute overlap and reduce-scatter backward.
+
+    Works with DistributedDataParallel (DDP) only (not DataParallel); pass
+    ``t_prime``/``bias`` to your optimizer like any other parameter
+    (reference ``README.md:19-20``).
+    """
+
+    def __init__(self, gpu_batch_size: int, strategy: str = "all_gather",
+                 col_chunk: Optional[int] = None, impl: str = "auto"):
+        super().__init__()
+        self.t_prime = nn.Parameter(torch.tensor(math.log(10.0)))
+        self.bias = nn.Parameter(torch.tensor(-10.0))
+        self.gpu_batch_size = gpu_batch_size
+        if strategy not in ("all_gather", "ring"):
+            raise ValueError(f"unknown strategy {strategy!r}")
+        self.strategy = strategy
+        self.col_chunk = col_chunk
+        self.impl = impl
+
+    def forward(self, image_embeddings: torch.Tensor,
+                text_embeddings: torch.Tensor, group=None) -> torch.Tensor:
+        world, rank = _world_and_rank(group)
+        b_txt = text_embeddings.shape[0]
+        if self.strategy == "all_gather" or world == 1:
+            all_txt = all_gather_with_grad(text_embeddings, group=group)
+            total = sigmoid_contrastive_loss(
+                image_embeddings, all_txt, self.t_prime, self.bias,
+                diag_offset=rank * b_txt, col_chunk=self.col_chunk,
+                impl=self.impl)
+        else:
+            total = _RingAllGatherLoss.apply(
+                image_embeddings, text_embeddings, self.t_prime, self.bias,
+                group, self.col_chunk, self.impl)
+        return total / self.gpu_batch_size
+
+
+class SigLipLoss(nn.Module):
+    """Ring-strategy SigLIP loss with caller-owned scale/bias parameters.
+
+    Parity with the open_clip-style reference (``rwightman_sigmoid_loss.py``):
+    local positive block first (``:69``), then ``W−1`` ring hops —
+    bidirectional pairs plus one unidirectional remainder when ``bidir``
+    (``:77-107``), else a pure unidirectional ring (``:108-122``) — each
+    remote chunk contributing negatives-only loss, every chunk normalized by
+    the local batch size (``:65``).  Gradients hop home through the
+    autograd-reversed exchanges.
+
+    ``use_horovod`` is unsupported, as in the reference (``:35``).
+    """
+
+    def __init__(self, cache_labels: bool = False, rank: int = 0,
+                 world_size: int = 1, bidir: bool = True,
+                 use_horovod: bool = False, impl: str = "auto",
+                 col_chunk: Optional[int] = None):
+        super().__init__()
+        if use_horovod:
+            raise NotImplementedError("horovod is not supported")
+        self.cache_labels = cache_labels  # kept for API parity; predicate labels need no cache
+        self.rank = rank
+        self.world_size = world_size
+        self.bidir = bidir
+        self.use_horovod = use_horovod
+        self.impl = impl
+        self.col_chunk = col_chunk
+
+    def _loss(self, image_features, text_features, logit_scale, logit_bias,
+              negative_only=False):
+        off = None if negative_only else 0
+        total = sigmoid_contrastive_loss(
+            image_features, text_features, logit_scale, logit_bias,
+            diag_offset=off, col_chunk=self.col_chunk, impl=self.impl)
+        return total / image_features.shape[0]
+
+    def forward(self, image_features, text_features, logit_scale, logit_bias,
+                output_dict: bool = False):
+        loss = self._loss(image_features, text_features, logit_scale,
+                          logit_bias)
+
+        if self.world_size > 1:
+            right_rank = (self.rank + 1) % self.world_size
+            left_rank = (self.rank - 1 + self.world_size) % self.world_size
+            if self.bidir:
+                to_right = to_left = text_features
+                num_bidir, remainder = divmod(self.world_size - 1, 2)
+                for _ in range(num_bidir):
+                    recv = neighbour_exchange_bidir_with_grad(
+                        left_rank, right_rank, to_left, to_right)
+                    for f in recv:
+                        loss = loss + self._loss(image_features, f,
+                                                 logit_scale, logit_bias,
+                                                 negative_only=True)
+                    to_left, to_right = recv
+                if remainder:
+                    recv = neighbour_exchange_with_grad(
+                        left_rank, right_rank, to_right)
+                    loss = loss + self._loss(image_features, recv,
+                                             logit_scale, logit_bias,
+                                             negative_only=True)
+            else:
+                to_right = text_features
+                for _ in range(self.world_size - 1):
+                    from_left = neighbour_exchange_with_grad(
+                        left_rank, right_rank, to_right)
+                    loss = loss + self._loss(image_features, from_left,
+                                             logit_scale, logit_bias,
+                                             negative_only=True)
+                    to_right = from_left
+
+        return {"contrastive_loss": loss} if output_dict else loss

@@ -1,0 +1,57 @@
+"""Scaling oracle: N-rank gradients == 1-rank gradients on the same global batch.
+
+Re-derivation of the reference's main correctness test
+(``test_distributed_sigmoid_loss.py:122-141``): the distributed loss under
+manual DDP grad averaging must reproduce the single-rank loss's encoder
+gradients at rtol=1e-3 — for both comm strategies.
+"""
+
+import pytest
+import torch
+
+from distributed_sigmoid_loss_amd import DistributedSigmoidLoss
+from distributed_sigmoid_loss_amd.parallel import average_gradients
+
+from helpers import encode_shard, run_distributed
+
+
+def ddp_step(rank, world, batch_per_rank, emb_dim, strategy, average):
+    img_enc, txt_enc, zi, zt = encode_shard(rank, world, batch_per_rank,
+                                            emb_dim)
+    loss_mod = DistributedSigmoidLoss(batch_per_rank, strategy=strategy)
+    loss = loss_mod(zi, zt)
+    loss.backward()
+    if average:
+        average_gradients(img_enc)
+        average_gradients(txt_enc)
+        average_gradients(loss_mod)
+    if rank == 0:
+        return {
+            "img": img_enc.weight.grad.clone(),
+            "txt": txt_enc.weight.grad.clone(),
+            "t_prime": loss_mod.t_prime.grad.clone(),
+            "bias": loss_mod.bias.grad.clone(),
+            "loss": loss.detach().clone(),
+        }
+
+
+@pytest.mark.parametrize("strategy", ["all_gather", "ring"])
+@pytest.mark.parametrize("world,batch", [(2, 4), (3, 3)])
+@pytest.mark.parametrize("emb_dim", [8, 64])
+def test_n_rank_matches_single_rank(strategy, world, batch, emb_dim):
+    multi = run_distributed(ddp_step, world, batch, emb_dim, strategy, True)[0]
+    single = run_distributed(ddp_step, 1, world * batch, emb_dim, strategy,
+                             True)[0]
+    for key in ("img", "txt", "t_prime", "bias"):
+        assert torch.allclose(multi[key], single[key], rtol=1e-3, atol=1e-6), \
+            f"{key} grads diverge: {multi[key]} vs {single[key]}"
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_ring_matches_all_gather_raw_grads(world):
+    """Both strategies of DistributedSigmoidLoss must be numerically
+    interchangeable at the raw (un-averaged) gradient level."""
+    ring = run_distributed(ddp_step, world, 4, 16, "ring", False)[0]
+    gather = run_distributed(ddp_step, world, 4, 16, "all_gather", False)[0]
+    for key in ("img", "txt", "t_prime", "bias", "loss"):
+        assert torch.allclose(ring[key], gather[key], rtol=1e-4, atol=1e-7), key

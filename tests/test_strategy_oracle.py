@@ -1,0 +1,82 @@
+"""Strategy-equivalence oracle: module-owned all-gather loss vs caller-owned
+ring SigLipLoss produce identical raw encoder gradients.
+
+Re-derivation of the reference's cross-implementation oracle
+(``test_sigmoid_loss_variants.py:93-113``): grad averaging deliberately
+skipped — raw rank-0 grads already agree because both losses route full
+cross-rank gradients through their differentiable comm.
+"""
+
+import math
+
+import pytest
+import torch
+import torch.nn as nn
+
+from distributed_sigmoid_loss_amd import DistributedSigmoidLoss, SigLipLoss
+
+from helpers import encode_shard, run_distributed
+
+
+def rw_step(rank, world, batch_per_rank, emb_dim, bidir):
+    img_enc, txt_enc, zi, zt = encode_shard(rank, world, batch_per_rank,
+                                            emb_dim)
+    logit_scale = nn.Parameter(torch.tensor(math.log(10.0)))
+    logit_bias = nn.Parameter(torch.tensor(-10.0))
+    loss_mod = SigLipLoss(rank=rank, world_size=world, bidir=bidir)
+    loss = loss_mod(zi, zt, logit_scale, logit_bias)
+    loss.backward()
+    if rank == 0:
+        return {
+            "img": img_enc.weight.grad.clone(),
+            "txt": txt_enc.weight.grad.clone(),
+            "scale": logit_scale.grad.clone(),
+            "bias": logit_bias.grad.clone(),
+            "loss": loss.detach().clone(),
+        }
+
+
+def ddp_step(rank, world, batch_per_rank, emb_dim, strategy):
+    img_enc, txt_enc, zi, zt = encode_shard(rank, world, batch_per_rank,
+                                            emb_dim)
+    loss_mod = DistributedSigmoidLoss(batch_per_rank, strategy=strategy)
+    loss = loss_mod(zi, zt)
+    loss.backward()
+    if rank == 0:
+        return {
+            "img": img_enc.weight.grad.clone(),
+            "txt": txt_enc.weight.grad.clone(),
+            "scale": loss_mod.t_prime.grad.clone(),
+            "bias": loss_mod.bias.grad.clone(),
+            "loss": loss.detach().clone(),
+        }
+
+
+@pytest.mark.parametrize("world,batch,dim", [(2, 4, 8), (3, 3, 8), (2, 4, 64)])
+@pytest.mark.parametrize("bidir", [True, False])
+@pytest.mark.parametrize("strategy", ["all_gather", "ring"])
+def test_siglip_matches_ddp(world, batch, dim, bidir, strategy):
+    rw = run_distributed(rw_step, world, batch, dim, bidir)[0]
+    ddp = run_distributed(ddp_step, world, batch, dim, strategy)[0]
+    for key in ("img", "txt", "scale", "bias", "loss"):
+        assert torch.allclose(rw[key], ddp[key], rtol=1e-3, atol=1e-6), \
+            f"{key}: {rw[key]} vs {ddp[key]}"
+
+
+def test_output_dict_form():
+    """SigLipLoss(output_dict=True) returns {'contrastive_loss': loss}
+    (reference rwightman_sigmoid_loss.py:124)."""
+    torch.manual_seed(0)
+    zi = torch.nn.functional.normalize(torch.randn(4, 8), dim=-1)
+    zt = torch.nn.functional.normalize(torch.randn(4, 8), dim=-1)
+    scale = torch.tensor(math.log(10.0))
+    bias = torch.tensor(-10.0)
+    mod = SigLipLoss(rank=0, world_size=1)
+    out = mod(zi, zt, scale, bias, output_dict=True)
+    assert set(out.keys()) == {"contrastive_loss"}
+    assert torch.allclose(out["contrastive_loss"], mod(zi, zt, scale, bias))
+
+
+def test_horovod_unsupported():
+    with pytest.raises(NotImplementedError):
+        SigLipLoss(use_horovod=True)
