@@ -119,33 +119,32 @@ class _RingAllGatherLoss(torch.autograd.Function):
         zimg, t_prime, bias = ctx.saved_tensors[:3]
         chunks = ctx.saved_tensors[3:]
         world, rank = ctx.world, ctx.rank
+        b = ctx.b_txt
 
-        dzimg = None
-        dt_prime = None
-        dbias = None
-        contribs = []
-        for s in range(world):
-            off = 0 if s == rank else None
-            dzi, dzt, dtp, dbi = chunk_loss_bwd(
-                zimg, chunks[s], t_prime, bias, off, grad_output,
-                col_chunk=ctx.col_chunk, impl=ctx.impl)
-            dzimg = dzi if dzimg is None else dzimg + dzi
-            dt_prime = dtp if dt_prime is None else dt_prime + dtp
-            dbias = dbi if dbias is None else dbias + dbi
-            contribs.append(dzt)
+        # One fused backward over the concatenated (W·b, d) text block —
+        # identical math to per-chunk calls (diag at the own-rank block),
+        # but a single dzimg accumulator and one slab loop.
+        all_txt = chunks[0] if world == 1 else torch.cat(chunks, dim=0)
+        dzimg, dtxt_flat, dt_prime, dbias = chunk_loss_bwd(
+            zimg, all_txt, t_prime, bias, rank * b, grad_output,
+            col_chunk=ctx.col_chunk, impl=ctx.impl)
 
         if world > 1:
-            flat = torch.cat(contribs, dim=0)
-            b = ctx.b_txt
+            # Each rank holds grad contributions for every rank's text shard;
+            # one reduce-scatter(SUM) delivers each shard's total home — the
+            # collapse of the reference's W−1 reversed ring hops
+            # (distributed_utils.py:74-77) into a single RCCL collective.
             if _backend_is_gloo(ctx.group):
-                dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=ctx.group)
-                dztxt = flat[rank * b:(rank + 1) * b].clone()
+                dist.all_reduce(dtxt_flat, op=dist.ReduceOp.SUM,
+                                group=ctx.group)
+                dztxt = dtxt_flat[rank * b:(rank + 1) * b].clone()
             else:
-                dztxt = torch.empty_like(contribs[rank])
-                dist.reduce_scatter_tensor(dztxt, flat, op=dist.ReduceOp.SUM,
+                dztxt = torch.empty_like(dtxt_flat[:b])
+                dist.reduce_scatter_tensor(dztxt, dtxt_flat,
+                                           op=dist.ReduceOp.SUM,
                                            group=ctx.group)
         else:
-            dztxt = contribs[0]
+            dztxt = dtxt_flat
 
         return dzimg, dztxt, dt_prime, dbias, None, None, None
 

@@ -1,0 +1,110 @@
+#!/usr/bin/env python3
+"""Per-phase timing decomposition of the flagship step on one GPU.
+
+Times each component with CUDA events: fwd kernel, bwd g-kernel, the two
+gradient GEMMs, encoder fwd/bwd, optimizer.  Run on a GPU box:
+    python tools/perf_probe.py [--batch 32768] [--dim 768]
+"""
+
+from __future__ import annotations
+
+import argparse
+import ctypes
+
+import torch
+import torch.nn.functional as F
+
+from distributed_sigmoid_loss_amd import ops, DistributedSigmoidLoss
+from distributed_sigmoid_loss_amd.models import TwoTowerModel
+
+
+def time_fn(fn, iters=10, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    start = torch.cuda.Event(enable_timing=True)
+    end = torch.cuda.Event(enable_timing=True)
+    start.record()
+    for _ in range(iters):
+        fn()
+    end.record()
+    torch.cuda.synchronize()
+    return start.elapsed_time(end) / iters
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--batch", type=int, default=32768)
+    p.add_argument("--dim", type=int, default=768)
+    p.add_argument("--col-chunk", type=int, default=8192)
+    args = p.parse_args()
+    b, d = args.batch, args.dim
+
+    torch.manual_seed(0)
+    dev = "cuda"
+    zi = F.normalize(torch.randn(b, d, device=dev), dim=-1).bfloat16()
+    zt = F.normalize(torch.randn(b, d, device=dev), dim=-1).bfloat16()
+    tp = torch.tensor(2.302585, device=dev)
+    bs = torch.tensor(-10.0, device=dev)
+    go = torch.tensor(1.0, device=dev)
+
+    flops_fwd = 2.0 * b * b * d
+
+    t_fwd = time_fn(lambda: ops.siglip_fwd(zi, zt, tp, bs, 0))
+    print(f"fwd fused kernel      : {t_fwd:8.3f} ms   "
+          f"{flops_fwd / t_fwd / 1e9:7.1f} TF/s")
+
+    # bwd pieces: g-kernel alone, then GEMMs alone.
+    lib = ops._require_lib()
+    c = min(args.col_chunk, b)
+    g = torch.empty((b, c), device=dev, dtype=torch.bfloat16)
+    scal = torch.zeros(2, device=dev, dtype=torch.float32)
+    ztc = zt[:c].contiguous()
+    stream = torch.cuda.current_stream().cuda_stream
+
+    def g_kernel():
+        ops._check(lib.siglip_bwd_g_bf16(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(zi.data_ptr()), ctypes.c_void_p(ztc.data_ptr()),
+            ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bs.data_ptr()),
+            ctypes.c_void_p(g.data_ptr()), ctypes.c_void_p(scal.data_ptr()),
+            b, c, d, 0), "bwd_g")
+
+    t_g = time_fn(g_kernel)
+    nchunks = (b + c - 1) // c
+    print(f"bwd g-kernel ({nchunks}x{c:6d}): {t_g * nchunks:8.3f} ms   "
+          f"{2.0 * b * c * d / t_g / 1e9:7.1f} TF/s per-chunk")
+
+    t_mm1 = time_fn(lambda: g @ ztc)
+    t_mm2 = time_fn(lambda: g.T @ zi)
+    print(f"GEMM g@zt   ({nchunks} chunks): {t_mm1 * nchunks:8.3f} ms   "
+          f"{2.0 * b * c * d / t_mm1 / 1e9:7.1f} TF/s")
+    print(f"GEMM gT@zi  ({nchunks} chunks): {t_mm2 * nchunks:8.3f} ms   "
+          f"{2.0 * b * c * d / t_mm2 / 1e9:7.1f} TF/s")
+
+    t_bwd = time_fn(lambda: ops.siglip_bwd(zi, zt, tp, bs, 0, go,
+                                           args.col_chunk))
+    print(f"bwd total (wrapper)   : {t_bwd:8.3f} ms")
+
+    # Whole training step as the bench runs it.
+    model = TwoTowerModel(d, d).to(device=dev, dtype=torch.bfloat16)
+    loss_mod = DistributedSigmoidLoss(b, col_chunk=args.col_chunk).cuda()
+    opt = torch.optim.SGD(list(model.parameters()) +
+                          list(loss_mod.parameters()), lr=1e-4)
+    img = torch.randn(b, d, device=dev, dtype=torch.bfloat16)
+    txt = torch.randn(b, d, device=dev, dtype=torch.bfloat16)
+
+    def full_step():
+        opt.zero_grad(set_to_none=True)
+        a, t_ = model(img, txt)
+        loss = loss_mod(a, t_)
+        loss.backward()
+        opt.step()
+
+    t_step = time_fn(full_step, iters=5, warmup=2)
+    print(f"full train step       : {t_step:8.3f} ms   "
+          f"({b / t_step * 1000:,.0f} pairs/s)")
+
+
+if __name__ == "__main__":
+    main()
