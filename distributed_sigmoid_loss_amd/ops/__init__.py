@@ -27,6 +27,12 @@ from .build import SO_PATH, build as build_extension
 
 _DIAG_NONE = -(2 ** 31)
 
+
+def _kernel_flags() -> int:
+    # bit 0: XCD-aware block remap (default on; SIGLIP_XCD_SWZ=0 disables
+    # for A/B profiling).
+    return 1 if os.environ.get("SIGLIP_XCD_SWZ", "1") != "0" else 0
+
 _lib = None
 _lib_err: Optional[str] = None
 
@@ -46,11 +52,10 @@ def _load():
         _lib_err = f"failed to load {SO_PATH}: {e}"
         return None
     lib.siglip_ext_abi.restype = ctypes.c_int
-    for fn in (lib.siglip_fwd_bf16,):
-        fn.restype = ctypes.c_int
-        fn.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] * 4
+    lib.siglip_fwd_bf16.restype = ctypes.c_int
+    lib.siglip_fwd_bf16.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] * 5
     lib.siglip_bwd_g_bf16.restype = ctypes.c_int
-    lib.siglip_bwd_g_bf16.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] * 4
+    lib.siglip_bwd_g_bf16.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] * 5
     _lib = lib
     return _lib
 
@@ -103,7 +108,8 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         ctypes.c_void_p(stream),
         ctypes.c_void_p(zimg.data_ptr()), ctypes.c_void_p(ztxt.data_ptr()),
         ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
-        ctypes.c_void_p(loss.data_ptr()), b, n, d, diag), "siglip_fwd_bf16")
+        ctypes.c_void_p(loss.data_ptr()), b, n, d, diag, _kernel_flags()),
+        "siglip_fwd_bf16")
     return loss
 
 
@@ -145,7 +151,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
             ctypes.c_void_p(zimg.data_ptr()), ctypes.c_void_p(zt.data_ptr()),
             ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
             ctypes.c_void_p(g.data_ptr()), ctypes.c_void_p(scal.data_ptr()),
-            b, c, d, diag), "siglip_bwd_g_bf16")
+            b, c, d, diag, _kernel_flags()), "siglip_bwd_g_bf16")
         dzimg_acc += (g @ zt).float()
         dztxt_acc[j0:j1] = (g.T @ zimg).float()
 

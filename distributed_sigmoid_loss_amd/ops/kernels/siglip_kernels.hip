@@ -16,16 +16,23 @@
 // the two rocBLAS GEMMs (dzimg = t·g@ztxt, dztxt = t·gᵀ@zimg) on the Python
 // side (ops/__init__.py).
 //
-// Kernel structure (one template, two epilogues):
-//   - 128×128 output tile per 256-thread (4-wave) workgroup,
-//     64×64 per wave as 4×4 fragments of v_mfma_f32_16x16x32_bf16.
-//   - K-loop BK=64, double-buffered LDS (write-after-barrier pipeline:
-//     global→reg loads issued before the MFMA phase, reg→LDS writes after).
-//   - LDS rows padded to 144 B (BK*2 + 16) so the 16-lane ds_read_b128
-//     fragment reads are bank-conflict-free (rows r: dword bank 36r mod 64,
-//     all 16 distinct slots) while keeping 16-B alignment.
-//   - fp32 accumulation throughout; scalar results via per-wave shuffle
-//     reduction + one atomicAdd per wave.
+// Kernel structure (one template; MODE 0 = fwd, MODE 1 = bwd-g):
+//   - 128×128 output tile per 256-thread (4-wave) workgroup, 64×64 per wave
+//     as 4×4 fragments of v_mfma_f32_16x16x32_bf16, fp32 accumulate.
+//   - K-loop BK=64, double-buffered LDS staged by global_load_lds_dwordx4
+//     (direct HBM→LDS DMA, no VGPR round trip); the DMA for tile k+1 is
+//     issued before the MFMA phase of tile k.
+//   - LDS image is lane-linear (glds requirement), so the bank swizzle is
+//     applied to the per-lane *source* address and the ds_read offset
+//     (both-sides rule): chunk' = chunk ^ ((row&7 + row>>3&1) & 7), which
+//     makes the 16-lane ds_read_b128 fragment reads bank-conflict-free.
+//   - Edge blocks (ragged b/n, d%64≠0) take a register-staged path writing
+//     the same swizzled LDS image with zero-fill guards; interior blocks
+//     (the entire grid at benchmark shapes) take the DMA fast path.
+//   - XCD-aware block remap (bijective): consecutive remapped ids share a
+//     ztxt panel inside one XCD's private L2.
+//   - MODE 1 stores the g tile through LDS (padded rows) so global writes
+//     are 16-B vectors instead of 2-B scatters.
 //
 // Requirements: d % 8 == 0 (16-byte K-vectors); b, n arbitrary (guarded).
 // Compile: hipcc --offload-arch=gfx950 -O3 -shared -fPIC.
@@ -36,81 +43,88 @@
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef const __attribute__((address_space(1))) unsigned int* gas_ptr;
+typedef __attribute__((address_space(3))) unsigned int* las_ptr;
 
 namespace {
 
-constexpr int BM = 128;        // image rows per block
-constexpr int BN = 128;        // text rows (logit cols) per block
-constexpr int BK = 64;         // K (emb dim) step
-constexpr int THREADS = 256;   // 4 waves
-constexpr int LDS_ROW = BK * 2 + 16;          // 144 B padded row
-constexpr int LDS_TILE = BM * LDS_ROW;        // 18432 B per operand tile
+constexpr int BM = 128;          // image rows per block
+constexpr int BN = 128;          // text rows (logit cols) per block
+constexpr int BK = 64;           // K (emb dim) step
+constexpr int THREADS = 256;     // 4 waves
+constexpr int ROW_BYTES = BK * 2;            // 128 B linear row
+constexpr int TILE_BYTES = BM * ROW_BYTES;   // 16 KiB per operand tile
 constexpr int DIAG_NONE = INT_MIN;
+constexpr int NXCD = 8;
+
+// g-tile staging rows padded to 272 B so the readback's 16-lane b128 groups
+// land on distinct bank slots.
+constexpr int G_ROW = BN * 2 + 16;
 
 __device__ __forceinline__ float softplus_f(float x) {
-  // log(1 + e^x), stable for all x.
   return fmaxf(x, 0.0f) + log1pf(__expf(-fabsf(x)));
 }
 
-struct StageRegs {
-  uint4 a[4];
-  uint4 b[4];
-};
+// Per-row XOR mask on the 16-B chunk index: distinct for the 8 even and 8
+// odd rows of any aligned 16-row group → conflict-free b128 fragment reads.
+__device__ __forceinline__ int kmask(int r) {
+  return ((r & 7) + ((r >> 3) & 1)) & 7;
+}
 
-// Issue the global loads for K-tile kt into registers (zero-filled outside
-// [b|n, d)).  Thread t covers row t>>1 of each tile, 32 K-elements starting at
-// (t&1)*32 — four 16-B vectors, coalesced along K.
-__device__ __forceinline__ void stage_load(
-    StageRegs& r, const __bf16* __restrict__ zimg,
-    const __bf16* __restrict__ ztxt, int row_base, int col_base, int b, int n,
-    int d, int k0) {
+// HBM→LDS DMA staging of one 128×64 bf16 tile (16 KiB).  Each of the 4 waves
+// issues 4 global_load_lds_dwordx4: LDS dest = wave-uniform base + lane*16
+// (lane-linear), per-lane source address carries the inverse swizzle.
+__device__ __forceinline__ void stage_glds(const __bf16* __restrict__ gsrc,
+                                           char* lds, int row0, int d,
+                                           int k0) {
+  const int w = threadIdx.x >> 6;
+  const int l = threadIdx.x & 63;
+  const int rsub = l >> 3;         // row within the 8-row group
+  const int c = l & 7;             // 16-B chunk within the row
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int rloc = (w * 4 + j) * 8 + rsub;
+    const int sc = c ^ kmask(rloc);
+    const __bf16* src = gsrc + (size_t)(row0 + rloc) * d + k0 + sc * 8;
+    // The LDS base is wave-uniform in value but threadIdx-derived, which the
+    // compiler treats as divergent — readfirstlane makes uniformity provable
+    // (avoids a waterfall loop around each DMA).
+    const int lbase = __builtin_amdgcn_readfirstlane((w * 4 + j) * 1024);
+    __builtin_amdgcn_global_load_lds((gas_ptr)src, (las_ptr)(lds + lbase),
+                                     16, 0, 0);
+  }
+}
+
+// Register-staged fallback for edge blocks: same swizzled LDS image,
+// zero-filled outside [rows, d).
+__device__ __forceinline__ void stage_guarded(const __bf16* __restrict__ gsrc,
+                                              char* lds, int row0, int rows,
+                                              int d, int k0) {
   const int t = threadIdx.x;
   const int row = t >> 1;
-  const int kh = (t & 1) * 32;
-  const int ga = row_base + row;
-  const int gb = col_base + row;
+  const int ch0 = (t & 1) * 4;
+  const int gr = row0 + row;
   const uint4 zero = {0u, 0u, 0u, 0u};
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
-    const int k = k0 + kh + i * 8;
-    r.a[i] = (ga < b && k < d)
-        ? *reinterpret_cast<const uint4*>(zimg + (size_t)ga * d + k) : zero;
-    r.b[i] = (gb < n && k < d)
-        ? *reinterpret_cast<const uint4*>(ztxt + (size_t)gb * d + k) : zero;
+    const int c = ch0 + i;
+    const int k = k0 + c * 8;
+    uint4 v = (gr < rows && k < d)
+        ? *reinterpret_cast<const uint4*>(gsrc + (size_t)gr * d + k) : zero;
+    *reinterpret_cast<uint4*>(lds + row * ROW_BYTES +
+                              ((c ^ kmask(row)) * 16)) = v;
   }
 }
 
-__device__ __forceinline__ void stage_write(const StageRegs& r, char* As,
-                                            char* Bs) {
-  const int t = threadIdx.x;
-  const int row = t >> 1;
-  const int off = row * LDS_ROW + (t & 1) * 64;
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    *reinterpret_cast<uint4*>(As + off + i * 16) = r.a[i];
-    *reinterpret_cast<uint4*>(Bs + off + i * 16) = r.b[i];
-  }
-}
-
-// MODE 0: forward loss.  MODE 1: backward g-slab + scalar partials.
-template <int MODE>
-__launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
+template <int MODE, bool INTERIOR>
+__device__ __forceinline__ void tile_body(
     const __bf16* __restrict__ zimg, const __bf16* __restrict__ ztxt,
-    const float* __restrict__ t_prime, const float* __restrict__ bias_p,
-    float* __restrict__ out,          // MODE 0: loss[1].  MODE 1: scal[2].
-    __bf16* __restrict__ g_out,       // MODE 1 only; leading dim = n
-    int b, int n, int d, int diag) {
-  __shared__ char smem[4 * LDS_TILE];
-
-  const int row_base = blockIdx.x * BM;
-  const int col_base = blockIdx.y * BN;
+    float t, float bias, float* __restrict__ out, __bf16* __restrict__ g_out,
+    int b, int n, int d, int diag, int row_base, int col_base, char* smem) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wrow = (wave >> 1) * 64;   // wave's 64×64 sub-tile origin
+  const int wrow = (wave >> 1) * 64;
   const int wcol = (wave & 1) * 64;
-
-  const float t = __expf(*t_prime);
-  const float bias = *bias_p;
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -119,34 +133,48 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int ktiles = (d + BK - 1) / BK;
-  StageRegs regs;
-  stage_load(regs, zimg, ztxt, row_base, col_base, b, n, d, 0);
-  stage_write(regs, smem, smem + LDS_TILE);
+
+  if (INTERIOR) {
+    stage_glds(zimg, smem, row_base, d, 0);
+    stage_glds(ztxt, smem + TILE_BYTES, col_base, d, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else {
+    stage_guarded(zimg, smem, row_base, b, d, 0);
+    stage_guarded(ztxt, smem + TILE_BYTES, col_base, n, d, 0);
+  }
   __syncthreads();
 
+  const int fr = lane & 15;
+  const int qbase = lane >> 4;          // K subgroup 0..3
   for (int kt = 0; kt < ktiles; ++kt) {
-    char* As = smem + (kt & 1) * (2 * LDS_TILE);
-    char* Bs = As + LDS_TILE;
-    const bool have_next = kt + 1 < ktiles;
-    if (have_next)
-      stage_load(regs, zimg, ztxt, row_base, col_base, b, n, d,
-                 (kt + 1) * BK);
-
-    // frag row/col within the wave tile: lane&15 selects the 16-row group
-    // element; lane>>4 selects the 8-wide K-subgroup.
-    const int fr = lane & 15;
-    const int fk = (lane >> 4) * 8;
+    char* As = smem + (kt & 1) * (2 * TILE_BYTES);
+    char* Bs = As + TILE_BYTES;
+    if (kt + 1 < ktiles) {
+      char* An = smem + ((kt + 1) & 1) * (2 * TILE_BYTES);
+      if (INTERIOR) {
+        stage_glds(zimg, An, row_base, d, (kt + 1) * BK);
+        stage_glds(ztxt, An + TILE_BYTES, col_base, d, (kt + 1) * BK);
+      } else {
+        stage_guarded(zimg, An, row_base, b, d, (kt + 1) * BK);
+        stage_guarded(ztxt, An + TILE_BYTES, col_base, n, d, (kt + 1) * BK);
+      }
+    }
 #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
+      const int q = kk * 4 + qbase;     // 16-B chunk index within the row
       bf16x8 afrag[4], bfrag[4];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < 4; ++mi) {
+        const int r = wrow + mi * 16 + fr;
         afrag[mi] = *reinterpret_cast<const bf16x8*>(
-            As + (wrow + mi * 16 + fr) * LDS_ROW + (kk * 32 + fk) * 2);
+            As + r * ROW_BYTES + ((q ^ kmask(r)) * 16));
+      }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
+      for (int ni = 0; ni < 4; ++ni) {
+        const int r = wcol + ni * 16 + fr;
         bfrag[ni] = *reinterpret_cast<const bf16x8*>(
-            Bs + (wcol + ni * 16 + fr) * LDS_ROW + (kk * 32 + fk) * 2);
+            Bs + r * ROW_BYTES + ((q ^ kmask(r)) * 16));
+      }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -154,17 +182,13 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
-
+    // The barrier both publishes the next tile's LDS writes and (glds path)
+    // drains the in-flight DMA via the compiler's vmcnt(0) before s_barrier.
     __syncthreads();
-    if (have_next) {
-      char* An = smem + ((kt + 1) & 1) * (2 * LDS_TILE);
-      stage_write(regs, An, An + LDS_TILE);
-      __syncthreads();
-    }
   }
 
-  // Epilogue.  C/D fragment layout of mfma_f32_16x16x32_bf16:
-  //   col = lane & 15, row = (lane >> 4) * 4 + reg.
+  // Epilogue.  C/D layout of mfma_f32_16x16x32_bf16:
+  //   col = lane&15, row = (lane>>4)*4 + reg.
   float s0 = 0.f, s1 = 0.f;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
@@ -172,19 +196,26 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     for (int ni = 0; ni < 4; ++ni) {
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int grow = row_base + wrow + mi * 16 + (lane >> 4) * 4 + reg;
-        const int gcol = col_base + wcol + ni * 16 + (lane & 15);
-        if (grow < b && gcol < n) {
+        const int lrow = wrow + mi * 16 + (lane >> 4) * 4 + reg;
+        const int lcol = wcol + ni * 16 + (lane & 15);
+        const int grow = row_base + lrow;
+        const int gcol = col_base + lcol;
+        if (INTERIOR || (grow < b && gcol < n)) {
           const float dot = acc[mi][ni][reg];
           const float z = dot * t + bias;
           const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
           if (MODE == 0) {
             s0 += softplus_f(pos ? -z : z);
           } else {
-            // g = -l * sigmoid(-l z) = -l / (1 + exp(l z))
             const float g = pos ? (-1.0f / (1.0f + __expf(z)))
                                 : (1.0f / (1.0f + __expf(-z)));
-            g_out[(size_t)grow * n + gcol] = (__bf16)g;
+            if (INTERIOR) {
+              // Stage through LDS (padded rows), vector-store below.
+              *reinterpret_cast<__bf16*>(smem + lrow * G_ROW + lcol * 2) =
+                  (__bf16)g;
+            } else {
+              g_out[(size_t)grow * n + gcol] = (__bf16)g;
+            }
             s0 += g * dot;
             s1 += g;
           }
@@ -193,7 +224,19 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     }
   }
 
-  // Wave-level tree reduction, one atomicAdd per wave per scalar.
+  if (MODE == 1 && INTERIOR) {
+    __syncthreads();
+    // 16-B vector stores: thread t writes 128 B of row t>>1.
+    const int row = threadIdx.x >> 1;
+    const int half = threadIdx.x & 1;
+    const char* src = smem + row * G_ROW + half * 128;
+    __bf16* dst = g_out + (size_t)(row_base + row) * n + col_base + half * 64;
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      reinterpret_cast<uint4*>(dst)[i] =
+          reinterpret_cast<const uint4*>(src)[i];
+  }
+
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) {
     s0 += __shfl_down(s0, off);
@@ -205,17 +248,57 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
   }
 }
 
+// MODE 0: forward loss.  MODE 1: backward g-slab + scalar partials.
+template <int MODE>
+__launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
+    const __bf16* __restrict__ zimg, const __bf16* __restrict__ ztxt,
+    const float* __restrict__ t_prime, const float* __restrict__ bias_p,
+    float* __restrict__ out,          // MODE 0: loss[1].  MODE 1: scal[2].
+    __bf16* __restrict__ g_out,       // MODE 1 only; leading dim = n
+    int b, int n, int d, int diag, int flags) {
+  __shared__ char smem[4 * TILE_BYTES];
+
+  // Bijective XCD-aware remap of the flat block id: each XCD gets a
+  // contiguous span, so neighbouring ids (sharing a ztxt panel) hit the
+  // same per-XCD L2.
+  int bx = blockIdx.x, by = blockIdx.y;
+  if (flags & 1) {
+    const int nwg = gridDim.x * gridDim.y;
+    const int id = by * gridDim.x + bx;
+    const int q = nwg / NXCD, r = nwg % NXCD;
+    const int xcd = id % NXCD, idx = id / NXCD;
+    const int nid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+        + idx;
+    bx = nid % gridDim.x;
+    by = nid / gridDim.x;
+  }
+  const int row_base = bx * BM;
+  const int col_base = by * BN;
+
+  const float t = __expf(*t_prime);
+  const float bias = *bias_p;
+
+  const bool interior = (row_base + BM <= b) && (col_base + BN <= n) &&
+      (d % BK == 0) && (n % 8 == 0);
+  if (interior)
+    tile_body<MODE, true>(zimg, ztxt, t, bias, out, g_out, b, n, d, diag,
+                          row_base, col_base, smem);
+  else
+    tile_body<MODE, false>(zimg, ztxt, t, bias, out, g_out, b, n, d, diag,
+                           row_base, col_base, smem);
+}
+
 inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
 }  // namespace
 
 extern "C" {
 
-int siglip_ext_abi(void) { return 1; }
+int siglip_ext_abi(void) { return 2; }
 
 int siglip_fwd_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                     const void* t_prime, const void* bias, void* loss_out,
-                    int b, int n, int d, int diag) {
+                    int b, int n, int d, int diag, int flags) {
   if (b <= 0 || n <= 0) return (int)hipSuccess;
   if (d % 8 != 0) return (int)hipErrorInvalidValue;
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
@@ -223,13 +306,14 @@ int siglip_fwd_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                      (hipStream_t)stream,
                      (const __bf16*)zimg, (const __bf16*)ztxt,
                      (const float*)t_prime, (const float*)bias,
-                     (float*)loss_out, (__bf16*)nullptr, b, n, d, diag);
+                     (float*)loss_out, (__bf16*)nullptr, b, n, d, diag,
+                     flags);
   return (int)hipGetLastError();
 }
 
 int siglip_bwd_g_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                       const void* t_prime, const void* bias, void* g_out,
-                      void* scal, int b, int n, int d, int diag) {
+                      void* scal, int b, int n, int d, int diag, int flags) {
   if (b <= 0 || n <= 0) return (int)hipSuccess;
   if (d % 8 != 0) return (int)hipErrorInvalidValue;
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
@@ -237,7 +321,7 @@ int siglip_bwd_g_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                      (hipStream_t)stream,
                      (const __bf16*)zimg, (const __bf16*)ztxt,
                      (const float*)t_prime, (const float*)bias,
-                     (float*)scal, (__bf16*)g_out, b, n, d, diag);
+                     (float*)scal, (__bf16*)g_out, b, n, d, diag, flags);
   return (int)hipGetLastError();
 }
 

@@ -10,6 +10,10 @@ from __future__ import annotations
 
 import argparse
 import ctypes
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
