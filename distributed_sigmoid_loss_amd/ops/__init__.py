@@ -29,9 +29,16 @@ _DIAG_NONE = -(2 ** 31)
 
 
 def _kernel_flags() -> int:
-    # bit 0: XCD-aware block remap (default on; SIGLIP_XCD_SWZ=0 disables
-    # for A/B profiling).
-    return 1 if os.environ.get("SIGLIP_XCD_SWZ", "1") != "0" else 0
+    # bit 0: XCD-contiguous block remap (default off — measured slightly
+    #        negative in the L3-resident regime; SIGLIP_XCD_SWZ=1 enables).
+    # bit 1: grouped block walk for L2 panel reuse (default on;
+    #        SIGLIP_GROUP_SWZ=0 disables for A/B profiling).
+    f = 0
+    if os.environ.get("SIGLIP_XCD_SWZ", "0") == "1":
+        f |= 1
+    if os.environ.get("SIGLIP_GROUP_SWZ", "1") != "0":
+        f |= 2
+    return f
 
 _lib = None
 _lib_err: Optional[str] = None

@@ -17,22 +17,25 @@
 // side (ops/__init__.py).
 //
 // Kernel structure (one template; MODE 0 = fwd, MODE 1 = bwd-g):
-//   - 128×128 output tile per 256-thread (4-wave) workgroup, 64×64 per wave
-//     as 4×4 fragments of v_mfma_f32_16x16x32_bf16, fp32 accumulate.
-//   - K-loop BK=64, double-buffered LDS staged by global_load_lds_dwordx4
-//     (direct HBM→LDS DMA, no VGPR round trip); the DMA for tile k+1 is
-//     issued before the MFMA phase of tile k.
+//   - 256×256 output tile per 512-thread (8-wave, 2×4) workgroup; each wave
+//     owns a 128×64 sub-tile as 8×4 fragments of v_mfma_f32_16x16x32_bf16,
+//     fp32 accumulate.  The 256² tile (vs 128²) halves the staged bytes per
+//     FLOP — this op is staging-bandwidth-bound at large batch.
+//   - K-loop BK=64, double-buffered LDS (4×32 KiB) staged by
+//     global_load_lds_dwordx4 (direct HBM→LDS DMA, no VGPR round trip); the
+//     DMA for tile k+1 is issued before the MFMA phase of tile k.
 //   - LDS image is lane-linear (glds requirement), so the bank swizzle is
 //     applied to the per-lane *source* address and the ds_read offset
-//     (both-sides rule): chunk' = chunk ^ ((row&7 + row>>3&1) & 7), which
-//     makes the 16-lane ds_read_b128 fragment reads bank-conflict-free.
+//     (both-sides rule): chunk' = chunk ^ ((row&7 + row>>3&1) & 7), making
+//     the 16-lane ds_read_b128 fragment reads bank-conflict-free.
 //   - Edge blocks (ragged b/n, d%64≠0) take a register-staged path writing
 //     the same swizzled LDS image with zero-fill guards; interior blocks
 //     (the entire grid at benchmark shapes) take the DMA fast path.
-//   - XCD-aware block remap (bijective): consecutive remapped ids share a
-//     ztxt panel inside one XCD's private L2.
-//   - MODE 1 stores the g tile through LDS (padded rows) so global writes
-//     are 16-B vectors instead of 2-B scatters.
+//   - Block-id remap for cache locality (flags): bit0 = XCD-contiguous
+//     remap, bit1 = grouped column-major walk (8 block-rows per group) so
+//     temporally-close blocks share operand panels in L2/L3.
+//   - MODE 1 stores the g tile through LDS (two 128-row passes, padded rows)
+//     so global writes are 16-B vectors instead of 2-B scatters.
 //
 // Requirements: d % 8 == 0 (16-byte K-vectors); b, n arbitrary (guarded).
 // Compile: hipcc --offload-arch=gfx950 -O3 -shared -fPIC.
@@ -48,17 +51,19 @@ typedef __attribute__((address_space(3))) unsigned int* las_ptr;
 
 namespace {
 
-constexpr int BM = 128;          // image rows per block
-constexpr int BN = 128;          // text rows (logit cols) per block
+constexpr int BM = 256;          // image rows per block
+constexpr int BN = 256;          // text rows (logit cols) per block
 constexpr int BK = 64;           // K (emb dim) step
-constexpr int THREADS = 256;     // 4 waves
+constexpr int THREADS = 512;     // 8 waves as 2(M)×4(N)
+constexpr int FM = 8;            // M fragments per wave (128 rows)
+constexpr int FN = 4;            // N fragments per wave (64 cols)
 constexpr int ROW_BYTES = BK * 2;            // 128 B linear row
-constexpr int TILE_BYTES = BM * ROW_BYTES;   // 16 KiB per operand tile
+constexpr int TILE_BYTES = BM * ROW_BYTES;   // 32 KiB per operand tile
 constexpr int DIAG_NONE = INT_MIN;
 constexpr int NXCD = 8;
+constexpr int GROUP_M = 8;       // block-rows per locality group (bit1)
 
-// g-tile staging rows padded to 272 B so the readback's 16-lane b128 groups
-// land on distinct bank slots.
+// g-tile epilogue staging: 128 rows of BN bf16, rows padded to 528 B.
 constexpr int G_ROW = BN * 2 + 16;
 
 __device__ __forceinline__ float softplus_f(float x) {
@@ -71,7 +76,7 @@ __device__ __forceinline__ int kmask(int r) {
   return ((r & 7) + ((r >> 3) & 1)) & 7;
 }
 
-// HBM→LDS DMA staging of one 128×64 bf16 tile (16 KiB).  Each of the 4 waves
+// HBM→LDS DMA staging of one 256×64 bf16 tile (32 KiB).  Each of the 8 waves
 // issues 4 global_load_lds_dwordx4: LDS dest = wave-uniform base + lane*16
 // (lane-linear), per-lane source address carries the inverse swizzle.
 __device__ __forceinline__ void stage_glds(const __bf16* __restrict__ gsrc,
@@ -101,7 +106,7 @@ __device__ __forceinline__ void stage_guarded(const __bf16* __restrict__ gsrc,
                                               char* lds, int row0, int rows,
                                               int d, int k0) {
   const int t = threadIdx.x;
-  const int row = t >> 1;
+  const int row = t >> 1;          // 0..255
   const int ch0 = (t & 1) * 4;
   const int gr = row0 + row;
   const uint4 zero = {0u, 0u, 0u, 0u};
@@ -123,14 +128,14 @@ __device__ __forceinline__ void tile_body(
     int b, int n, int d, int diag, int row_base, int col_base, char* smem) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wrow = (wave >> 1) * 64;
-  const int wcol = (wave & 1) * 64;
+  const int wrow = (wave >> 2) * 128;   // wave sub-tile origin: 2×4 grid
+  const int wcol = (wave & 3) * 64;
 
-  f32x4 acc[4][4];
+  f32x4 acc[FM][FN];
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi)
+  for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int ni = 0; ni < FN; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int ktiles = (d + BK - 1) / BK;
 
@@ -162,23 +167,23 @@ __device__ __forceinline__ void tile_body(
 #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
       const int q = kk * 4 + qbase;     // 16-B chunk index within the row
-      bf16x8 afrag[4], bfrag[4];
+      bf16x8 afrag[FM], bfrag[FN];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
+      for (int mi = 0; mi < FM; ++mi) {
         const int r = wrow + mi * 16 + fr;
         afrag[mi] = *reinterpret_cast<const bf16x8*>(
             As + r * ROW_BYTES + ((q ^ kmask(r)) * 16));
       }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
+      for (int ni = 0; ni < FN; ++ni) {
         const int r = wcol + ni * 16 + fr;
         bfrag[ni] = *reinterpret_cast<const bf16x8*>(
             Bs + r * ROW_BYTES + ((q ^ kmask(r)) * 16));
       }
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < FN; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
@@ -190,51 +195,74 @@ __device__ __forceinline__ void tile_body(
   // Epilogue.  C/D layout of mfma_f32_16x16x32_bf16:
   //   col = lane&15, row = (lane>>4)*4 + reg.
   float s0 = 0.f, s1 = 0.f;
+  if (MODE == 1 && INTERIOR) {
+    // Two 128-row passes: waves with wrow==pass*128 stage bf16 g into LDS,
+    // then all 512 threads issue 16-B vector stores of that half.
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+    for (int pass = 0; pass < 2; ++pass) {
+      if (wrow == pass * 128) {
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
+        for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const int lrow = wrow + mi * 16 + (lane >> 4) * 4 + reg;
-        const int lcol = wcol + ni * 16 + (lane & 15);
-        const int grow = row_base + lrow;
-        const int gcol = col_base + lcol;
-        if (INTERIOR || (grow < b && gcol < n)) {
-          const float dot = acc[mi][ni][reg];
-          const float z = dot * t + bias;
-          const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
-          if (MODE == 0) {
-            s0 += softplus_f(pos ? -z : z);
-          } else {
-            const float g = pos ? (-1.0f / (1.0f + __expf(z)))
-                                : (1.0f / (1.0f + __expf(-z)));
-            if (INTERIOR) {
-              // Stage through LDS (padded rows), vector-store below.
+          for (int ni = 0; ni < FN; ++ni) {
+#pragma unroll
+            for (int reg = 0; reg < 4; ++reg) {
+              const int lrow = (wrow & 127) + mi * 16 + (lane >> 4) * 4 + reg;
+              const int lcol = wcol + ni * 16 + (lane & 15);
+              const float dot = acc[mi][ni][reg];
+              const float z = dot * t + bias;
+              const bool pos = (diag != DIAG_NONE) &&
+                  (col_base + lcol == row_base + pass * 128 + lrow + diag);
+              const float g = pos ? (-1.0f / (1.0f + __expf(z)))
+                                  : (1.0f / (1.0f + __expf(-z)));
               *reinterpret_cast<__bf16*>(smem + lrow * G_ROW + lcol * 2) =
                   (__bf16)g;
-            } else {
-              g_out[(size_t)grow * n + gcol] = (__bf16)g;
+              s0 += g * dot;
+              s1 += g;
             }
-            s0 += g * dot;
-            s1 += g;
+          }
+        }
+      }
+      __syncthreads();
+      {
+        const int row = threadIdx.x >> 2;        // 128 rows, 4 threads/row
+        const int qq = threadIdx.x & 3;
+        const char* src = smem + row * G_ROW + qq * 128;
+        __bf16* dst = g_out +
+            (size_t)(row_base + pass * 128 + row) * n + col_base + qq * 64;
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          reinterpret_cast<uint4*>(dst)[i] =
+              reinterpret_cast<const uint4*>(src)[i];
+      }
+      __syncthreads();
+    }
+  } else {
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int grow = row_base + wrow + mi * 16 + (lane >> 4) * 4 + reg;
+          const int gcol = col_base + wcol + ni * 16 + (lane & 15);
+          if (INTERIOR || (grow < b && gcol < n)) {
+            const float dot = acc[mi][ni][reg];
+            const float z = dot * t + bias;
+            const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
+            if (MODE == 0) {
+              s0 += softplus_f(pos ? -z : z);
+            } else {
+              const float g = pos ? (-1.0f / (1.0f + __expf(z)))
+                                  : (1.0f / (1.0f + __expf(-z)));
+              g_out[(size_t)grow * n + gcol] = (__bf16)g;
+              s0 += g * dot;
+              s1 += g;
+            }
           }
         }
       }
     }
-  }
-
-  if (MODE == 1 && INTERIOR) {
-    __syncthreads();
-    // 16-B vector stores: thread t writes 128 B of row t>>1.
-    const int row = threadIdx.x >> 1;
-    const int half = threadIdx.x & 1;
-    const char* src = smem + row * G_ROW + half * 128;
-    __bf16* dst = g_out + (size_t)(row_base + row) * n + col_base + half * 64;
-#pragma unroll
-    for (int i = 0; i < 8; ++i)
-      reinterpret_cast<uint4*>(dst)[i] =
-          reinterpret_cast<const uint4*>(src)[i];
   }
 
 #pragma unroll
@@ -258,19 +286,27 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     int b, int n, int d, int diag, int flags) {
   __shared__ char smem[4 * TILE_BYTES];
 
-  // Bijective XCD-aware remap of the flat block id: each XCD gets a
-  // contiguous span, so neighbouring ids (sharing a ztxt panel) hit the
-  // same per-XCD L2.
-  int bx = blockIdx.x, by = blockIdx.y;
+  const int gx = gridDim.x, gy = gridDim.y;
+  int id = blockIdx.y * gx + blockIdx.x;
   if (flags & 1) {
-    const int nwg = gridDim.x * gridDim.y;
-    const int id = by * gridDim.x + bx;
+    // Bijective XCD-contiguous remap: each XCD gets a contiguous id span.
+    const int nwg = gx * gy;
     const int q = nwg / NXCD, r = nwg % NXCD;
     const int xcd = id % NXCD, idx = id / NXCD;
-    const int nid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
-        + idx;
-    bx = nid % gridDim.x;
-    by = nid / gridDim.x;
+    id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  int bx, by;
+  if (flags & 2) {
+    // Grouped column-major walk: GROUP_M block-rows per group, so
+    // temporally-close blocks reuse both operand panels.
+    const int group = id / (GROUP_M * gy);
+    const int within = id % (GROUP_M * gy);
+    const int gm = min(GROUP_M, gx - group * GROUP_M);
+    bx = group * GROUP_M + within % gm;
+    by = within / gm;
+  } else {
+    bx = id % gx;
+    by = id / gx;
   }
   const int row_base = bx * BM;
   const int col_base = by * BN;
@@ -294,7 +330,7 @@ inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
 extern "C" {
 
-int siglip_ext_abi(void) { return 2; }
+int siglip_ext_abi(void) { return 3; }
 
 int siglip_fwd_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                     const void* t_prime, const void* bias, void* loss_out,

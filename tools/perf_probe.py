@@ -72,7 +72,7 @@ def main():
             ctypes.c_void_p(zi.data_ptr()), ctypes.c_void_p(ztc.data_ptr()),
             ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bs.data_ptr()),
             ctypes.c_void_p(g.data_ptr()), ctypes.c_void_p(scal.data_ptr()),
-            b, c, d, 0), "bwd_g")
+            b, c, d, 0, ops._kernel_flags()), "bwd_g")
 
     t_g = time_fn(g_kernel)
     nchunks = (b + c - 1) // c
