@@ -138,32 +138,10 @@ __device__ __forceinline__ void tile_body(
     for (int ni = 0; ni < FN; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int ktiles = (d + BK - 1) / BK;
-
-  if (INTERIOR) {
-    stage_glds(zimg, smem, row_base, d, 0);
-    stage_glds(ztxt, smem + TILE_BYTES, col_base, d, 0);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  } else {
-    stage_guarded(zimg, smem, row_base, b, d, 0);
-    stage_guarded(ztxt, smem + TILE_BYTES, col_base, n, d, 0);
-  }
-  __syncthreads();
-
   const int fr = lane & 15;
   const int qbase = lane >> 4;          // K subgroup 0..3
-  for (int kt = 0; kt < ktiles; ++kt) {
-    char* As = smem + (kt & 1) * (2 * TILE_BYTES);
-    char* Bs = As + TILE_BYTES;
-    if (kt + 1 < ktiles) {
-      char* An = smem + ((kt + 1) & 1) * (2 * TILE_BYTES);
-      if (INTERIOR) {
-        stage_glds(zimg, An, row_base, d, (kt + 1) * BK);
-        stage_glds(ztxt, An + TILE_BYTES, col_base, d, (kt + 1) * BK);
-      } else {
-        stage_guarded(zimg, An, row_base, b, d, (kt + 1) * BK);
-        stage_guarded(ztxt, An + TILE_BYTES, col_base, n, d, (kt + 1) * BK);
-      }
-    }
+
+  auto compute_ktile = [&](char* As, char* Bs) {
 #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
       const int q = kk * 4 + qbase;     // 16-B chunk index within the row
@@ -187,9 +165,52 @@ __device__ __forceinline__ void tile_body(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
     }
-    // The barrier both publishes the next tile's LDS writes and (glds path)
-    // drains the in-flight DMA via the compiler's vmcnt(0) before s_barrier.
+  };
+
+  if (INTERIOR) {
+    // Counted-vmcnt pipeline: the DMA for tile k+1 stays in flight across
+    // the barriers and the whole MFMA phase of tile k (a __syncthreads here
+    // would emit s_waitcnt vmcnt(0) and drain it — use raw s_barrier).
+    // Each wave issues 8 global_load_lds per tile (4 per operand), so the
+    // wait for "tile k landed, k+1 still flying" is vmcnt(8).
+    stage_glds(zimg, smem, row_base, d, 0);
+    stage_glds(ztxt, smem + TILE_BYTES, col_base, d, 0);
+    for (int kt = 0; kt < ktiles; ++kt) {
+      char* As = smem + (kt & 1) * (2 * TILE_BYTES);
+      char* Bs = As + TILE_BYTES;
+      if (kt + 1 < ktiles) {
+        // Safe to overwrite buf[(kt+1)&1]: the barrier ending iteration
+        // kt-1 fenced every wave's reads of tile kt-1 from it.
+        char* An = smem + ((kt + 1) & 1) * (2 * TILE_BYTES);
+        stage_glds(zimg, An, row_base, d, (kt + 1) * BK);
+        stage_glds(ztxt, An + TILE_BYTES, col_base, d, (kt + 1) * BK);
+        // Wait + barrier in ONE asm statement with a "memory" clobber: the
+        // plain s_barrier builtin is not a compiler memory fence, and the
+        // scheduler was observed hoisting the fragment ds_reads above it —
+        // reading rows another wave's DMA had not landed yet.
+        asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
+      }
+      compute_ktile(As, Bs);
+      // Fence reads of buf[kt&1] before the next iteration's DMA overwrites.
+      asm volatile("s_barrier" ::: "memory");
+    }
+  } else {
+    stage_guarded(zimg, smem, row_base, b, d, 0);
+    stage_guarded(ztxt, smem + TILE_BYTES, col_base, n, d, 0);
     __syncthreads();
+    for (int kt = 0; kt < ktiles; ++kt) {
+      char* As = smem + (kt & 1) * (2 * TILE_BYTES);
+      char* Bs = As + TILE_BYTES;
+      if (kt + 1 < ktiles) {
+        char* An = smem + ((kt + 1) & 1) * (2 * TILE_BYTES);
+        stage_guarded(zimg, An, row_base, b, d, (kt + 1) * BK);
+        stage_guarded(ztxt, An + TILE_BYTES, col_base, n, d, (kt + 1) * BK);
+      }
+      compute_ktile(As, Bs);
+      __syncthreads();
+    }
   }
 
   // Epilogue.  C/D layout of mfma_f32_16x16x32_bf16:
