@@ -58,6 +58,14 @@ def main():
     print(f"fwd fused kernel      : {t_fwd:8.3f} ms   "
           f"{flops_fwd / t_fwd / 1e9:7.1f} TF/s")
 
+    # rocBLAS ceiling reference: the same-shape plain GEMM (materializes the
+    # logits we fuse away) — an upper bound on achievable MFMA throughput.
+    ref_n = min(b, 8192)
+    zt_ref = zt[:ref_n].contiguous()
+    t_ref = time_fn(lambda: zi @ zt_ref.T)
+    print(f"rocBLAS zi@zt.T ref   : {t_ref:8.3f} ms   "
+          f"{2.0 * b * ref_n * d / t_ref / 1e9:7.1f} TF/s  (n={ref_n})")
+
     # bwd pieces: g-kernel alone, then GEMMs alone.
     lib = ops._require_lib()
     c = min(args.col_chunk, b)
