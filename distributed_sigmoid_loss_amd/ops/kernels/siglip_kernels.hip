@@ -66,14 +66,26 @@ constexpr int GROUP_M = 8;       // block-rows per locality group (bit1)
 // g-tile epilogue staging: 128 rows of BN bf16, rows padded to 528 B.
 constexpr int G_ROW = BN * 2 + 16;
 
+// softplus via the inlined fast log/exp (log1pf is a device-lib CALL —
+// measured as s_getpc/s_setpc pairs inside the epilogue).  For y = e^{-|x|}
+// ∈ (0,1] the naive log(1+y) loses only ~y·ε absolute accuracy near y→0,
+// where the term itself vanishes — fine at bf16-class tolerances.
 __device__ __forceinline__ float softplus_f(float x) {
-  return fmaxf(x, 0.0f) + log1pf(__expf(-fabsf(x)));
+  return fmaxf(x, 0.0f) + __logf(1.0f + __expf(-fabsf(x)));
 }
 
-// Per-row XOR mask on the 16-B chunk index: distinct for the 8 even and 8
-// odd rows of any aligned 16-row group → conflict-free b128 fragment reads.
+__device__ __forceinline__ float sigmoid_fast(float negz) {
+  // 1/(1+e^{negz}) via the single-instruction v_rcp_f32.
+  return __builtin_amdgcn_rcpf(1.0f + __expf(negz));
+}
+
+// Per-row XOR mask on the 16-B chunk index.  Depends only on r&15 (so
+// fragment reads at row = base + mi*16 + fr share one per-lane mask) and is
+// conflict-free for ds_read_b128 under both contiguous and interleaved
+// 16-lane servicing groups: same-parity rows differing by 2 always differ
+// in mask bit 2, and a fixed-q column read over 16 rows sees all 8 masks.
 __device__ __forceinline__ int kmask(int r) {
-  return ((r & 7) + ((r >> 3) & 1)) & 7;
+  return (((r >> 1) & 1) << 2) | ((r >> 2) & 3);
 }
 
 // HBM→LDS DMA staging of one 256×64 bf16 tile (32 KiB).  Each of the 8 waves
@@ -141,62 +153,119 @@ __device__ __forceinline__ void tile_body(
   const int fr = lane & 15;
   const int qbase = lane >> 4;          // K subgroup 0..3
 
-  auto compute_ktile = [&](char* As, char* Bs) {
-#pragma unroll
-    for (int kk = 0; kk < BK / 32; ++kk) {
-      const int q = kk * 4 + qbase;     // 16-B chunk index within the row
-      bf16x8 afrag[FM], bfrag[FN];
-#pragma unroll
-      for (int mi = 0; mi < FM; ++mi) {
-        const int r = wrow + mi * 16 + fr;
-        afrag[mi] = *reinterpret_cast<const bf16x8*>(
-            As + r * ROW_BYTES + ((q ^ kmask(r)) * 16));
-      }
-#pragma unroll
-      for (int ni = 0; ni < FN; ++ni) {
-        const int r = wcol + ni * 16 + fr;
-        bfrag[ni] = *reinterpret_cast<const bf16x8*>(
-            Bs + r * ROW_BYTES + ((q ^ kmask(r)) * 16));
-      }
-#pragma unroll
-      for (int mi = 0; mi < FM; ++mi)
-#pragma unroll
-        for (int ni = 0; ni < FN; ++ni)
-          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
-    }
-  };
-
   if (INTERIOR) {
-    // Counted-vmcnt pipeline: the DMA for tile k+1 stays in flight across
-    // the barriers and the whole MFMA phase of tile k (a __syncthreads here
-    // would emit s_waitcnt vmcnt(0) and drain it — use raw s_barrier).
-    // Each wave issues 8 global_load_lds per tile (4 per operand), so the
-    // wait for "tile k landed, k+1 still flying" is vmcnt(8).
-    stage_glds(zimg, smem, row_base, d, 0);
-    stage_glds(ztxt, smem + TILE_BYTES, col_base, d, 0);
+    // ---- Per-lane precomputed addressing (the naive form recomputed every
+    // glds source with a 64-bit multiply and every fragment address with
+    // 2-3 VALU ops per read — measured VALU-bound, VALUBusy ≈ 5×MfmaUtil).
+    const int w = threadIdx.x >> 6;
+    const int rsub = lane >> 3;          // glds: row within an 8-row group
+    const int cch = lane & 7;            // glds: 16-B chunk within the row
+    // glds source voffsets (bytes) for the 4 DMA issues of this wave, per
+    // operand; the K advance lives in the uniform base pointers below.
+    int va[4], vb[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int rloc = (w * 4 + j) * 8 + rsub;
+      const int sc = cch ^ kmask(rloc);
+      va[j] = rloc * d * 2 + sc * 16;
+      vb[j] = va[j];                     // same tile geometry for B
+    }
+    // Fragment ds_read addresses: chunk mask depends only on fr = r&15, so
+    // each (operand, kk) needs one per-lane base; mi/ni go into the 16-bit
+    // instruction offset (mi*2048 B).  Buffer double-toggle = one v_xor.
+    const int mk = kmask(fr);
+    int aAddr[2], bAddr[2];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int ch = (kk * 4 + qbase) ^ mk;
+      aAddr[kk] = (wrow + fr) * ROW_BYTES + ch * 16;
+      bAddr[kk] = TILE_BYTES + (wcol + fr) * ROW_BYTES + ch * 16;
+    }
+    const char* abase = reinterpret_cast<const char*>(zimg) +
+        (size_t)row_base * d * 2;
+    const char* bbase = reinterpret_cast<const char*>(ztxt) +
+        (size_t)col_base * d * 2;
+
+    auto stage = [&](int buf) {
+      const int lb = __builtin_amdgcn_readfirstlane((w * 4) * 1024) +
+          buf * (2 * TILE_BYTES);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        __builtin_amdgcn_global_load_lds(
+            (gas_ptr)(abase + va[j]), (las_ptr)(smem + lb + j * 1024),
+            16, 0, 0);
+        __builtin_amdgcn_global_load_lds(
+            (gas_ptr)(bbase + vb[j]),
+            (las_ptr)(smem + lb + TILE_BYTES + j * 1024), 16, 0, 0);
+      }
+      abase += BK * 2;                   // uniform K advance (SALU)
+      bbase += BK * 2;
+    };
+
+    stage(0);
     for (int kt = 0; kt < ktiles; ++kt) {
-      char* As = smem + (kt & 1) * (2 * TILE_BYTES);
-      char* Bs = As + TILE_BYTES;
       if (kt + 1 < ktiles) {
         // Safe to overwrite buf[(kt+1)&1]: the barrier ending iteration
         // kt-1 fenced every wave's reads of tile kt-1 from it.
-        char* An = smem + ((kt + 1) & 1) * (2 * TILE_BYTES);
-        stage_glds(zimg, An, row_base, d, (kt + 1) * BK);
-        stage_glds(ztxt, An + TILE_BYTES, col_base, d, (kt + 1) * BK);
+        stage((kt + 1) & 1);
         // Wait + barrier in ONE asm statement with a "memory" clobber: the
         // plain s_barrier builtin is not a compiler memory fence, and the
         // scheduler was observed hoisting the fragment ds_reads above it —
-        // reading rows another wave's DMA had not landed yet.
+        // reading rows another wave's DMA had not landed yet.  vmcnt(8):
+        // this wave's 8 DMAs for tile kt landed, tile kt+1's still flying.
         asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
       } else {
         asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
       }
-      compute_ktile(As, Bs);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bf16x8 afrag[FM], bfrag[FN];
+#pragma unroll
+        for (int mi = 0; mi < FM; ++mi)
+          afrag[mi] = *reinterpret_cast<const bf16x8*>(
+              smem + aAddr[kk] + mi * (16 * ROW_BYTES));
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni)
+          bfrag[ni] = *reinterpret_cast<const bf16x8*>(
+              smem + bAddr[kk] + ni * (16 * ROW_BYTES));
+#pragma unroll
+        for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < FN; ++ni)
+            acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+      }
+      aAddr[0] ^= 2 * TILE_BYTES; aAddr[1] ^= 2 * TILE_BYTES;
+      bAddr[0] ^= 2 * TILE_BYTES; bAddr[1] ^= 2 * TILE_BYTES;
       // Fence reads of buf[kt&1] before the next iteration's DMA overwrites.
       asm volatile("s_barrier" ::: "memory");
     }
   } else {
+    auto compute_ktile = [&](char* As, char* Bs) {
+#pragma unroll
+      for (int kk = 0; kk < BK / 32; ++kk) {
+        const int q = kk * 4 + qbase;   // 16-B chunk index within the row
+        bf16x8 afrag[FM], bfrag[FN];
+#pragma unroll
+        for (int mi = 0; mi < FM; ++mi) {
+          const int r = wrow + mi * 16 + fr;
+          afrag[mi] = *reinterpret_cast<const bf16x8*>(
+              As + r * ROW_BYTES + ((q ^ kmask(r)) * 16));
+        }
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni) {
+          const int r = wcol + ni * 16 + fr;
+          bfrag[ni] = *reinterpret_cast<const bf16x8*>(
+              Bs + r * ROW_BYTES + ((q ^ kmask(r)) * 16));
+        }
+#pragma unroll
+        for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < FN; ++ni)
+            acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+      }
+    };
     stage_guarded(zimg, smem, row_base, b, d, 0);
     stage_guarded(ztxt, smem + TILE_BYTES, col_base, n, d, 0);
     __syncthreads();
@@ -234,8 +303,7 @@ __device__ __forceinline__ void tile_body(
               const float z = dot * t + bias;
               const bool pos = (diag != DIAG_NONE) &&
                   (col_base + lcol == row_base + pass * 128 + lrow + diag);
-              const float g = pos ? (-1.0f / (1.0f + __expf(z)))
-                                  : (1.0f / (1.0f + __expf(-z)));
+              const float g = pos ? -sigmoid_fast(z) : sigmoid_fast(-z);
               *reinterpret_cast<__bf16*>(smem + lrow * G_ROW + lcol * 2) =
                   (__bf16)g;
               s0 += g * dot;
@@ -274,8 +342,7 @@ __device__ __forceinline__ void tile_body(
             if (MODE == 0) {
               s0 += softplus_f(pos ? -z : z);
             } else {
-              const float g = pos ? (-1.0f / (1.0f + __expf(z)))
-                                  : (1.0f / (1.0f + __expf(-z)));
+              const float g = pos ? -sigmoid_fast(z) : sigmoid_fast(-z);
               g_out[(size_t)grow * n + gcol] = (__bf16)g;
               s0 += g * dot;
               s1 += g;
