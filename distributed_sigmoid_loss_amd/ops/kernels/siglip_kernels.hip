@@ -285,48 +285,7 @@ __device__ __forceinline__ void tile_body(
   // Epilogue.  C/D layout of mfma_f32_16x16x32_bf16:
   //   col = lane&15, row = (lane>>4)*4 + reg.
   float s0 = 0.f, s1 = 0.f;
-  if (MODE == 1 && INTERIOR) {
-    // Two 128-row passes: waves with wrow==pass*128 stage bf16 g into LDS,
-    // then all 512 threads issue 16-B vector stores of that half.
-#pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
-      if (wrow == pass * 128) {
-#pragma unroll
-        for (int mi = 0; mi < FM; ++mi) {
-#pragma unroll
-          for (int ni = 0; ni < FN; ++ni) {
-#pragma unroll
-            for (int reg = 0; reg < 4; ++reg) {
-              const int lrow = (wrow & 127) + mi * 16 + (lane >> 4) * 4 + reg;
-              const int lcol = wcol + ni * 16 + (lane & 15);
-              const float dot = acc[mi][ni][reg];
-              const float z = dot * t + bias;
-              const bool pos = (diag != DIAG_NONE) &&
-                  (col_base + lcol == row_base + pass * 128 + lrow + diag);
-              const float g = pos ? -sigmoid_fast(z) : sigmoid_fast(-z);
-              *reinterpret_cast<__bf16*>(smem + lrow * G_ROW + lcol * 2) =
-                  (__bf16)g;
-              s0 += g * dot;
-              s1 += g;
-            }
-          }
-        }
-      }
-      __syncthreads();
-      {
-        const int row = threadIdx.x >> 2;        // 128 rows, 4 threads/row
-        const int qq = threadIdx.x & 3;
-        const char* src = smem + row * G_ROW + qq * 128;
-        __bf16* dst = g_out +
-            (size_t)(row_base + pass * 128 + row) * n + col_base + qq * 64;
-#pragma unroll
-        for (int i = 0; i < 8; ++i)
-          reinterpret_cast<uint4*>(dst)[i] =
-              reinterpret_cast<const uint4*>(src)[i];
-      }
-      __syncthreads();
-    }
-  } else {
+  if (MODE == 0) {
 #pragma unroll
     for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
@@ -336,20 +295,44 @@ __device__ __forceinline__ void tile_body(
           const int grow = row_base + wrow + mi * 16 + (lane >> 4) * 4 + reg;
           const int gcol = col_base + wcol + ni * 16 + (lane & 15);
           if (INTERIOR || (grow < b && gcol < n)) {
-            const float dot = acc[mi][ni][reg];
-            const float z = dot * t + bias;
+            const float z = acc[mi][ni][reg] * t + bias;
             const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
-            if (MODE == 0) {
-              s0 += softplus_f(pos ? -z : z);
-            } else {
-              const float g = pos ? -sigmoid_fast(z) : sigmoid_fast(-z);
-              g_out[(size_t)grow * n + gcol] = (__bf16)g;
-              s0 += g * dot;
-              s1 += g;
-            }
+            s0 += softplus_f(pos ? -z : z);
           }
         }
       }
+    }
+  } else {
+    // One per-lane base offset + per-(mi,reg) scalar row offset keeps the
+    // store addressing affine — per-element (size_t)grow*n math made the
+    // allocator hoist 128 addresses and spill.  Caller guarantees
+    // b*n*2 < 2^32 (ops/__init__.py column-chunks the slab).
+    __bf16* gb = g_out + (size_t)row_base * n + col_base;
+    const unsigned lane_off =
+        (unsigned)(wrow + (lane >> 4) * 4) * (unsigned)n
+        + (unsigned)(wcol + (lane & 15));
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)n;
+        const int grow = row_base + wrow + mi * 16 + (lane >> 4) * 4 + reg;
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni) {
+          const int gcol = col_base + wcol + ni * 16 + (lane & 15);
+          if (INTERIOR || (grow < b && gcol < n)) {
+            const float dot = acc[mi][ni][reg];
+            const float z = dot * t + bias;
+            const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
+            const float gv = sigmoid_fast(pos ? z : -z);
+            const float g = pos ? -gv : gv;
+            gb[lane_off + row_off + ni * 16] = (__bf16)g;
+            s0 += g * dot;
+            s1 += g;
+          }
+        }
+      }
+      __builtin_amdgcn_sched_barrier(0);  // cap epilogue register pressure
     }
   }
 
@@ -364,29 +347,16 @@ __device__ __forceinline__ void tile_body(
   }
 }
 
-// MODE 0: forward loss.  MODE 1: backward g-slab + scalar partials.
-template <int MODE>
-__launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
-    const __bf16* __restrict__ zimg, const __bf16* __restrict__ ztxt,
-    const float* __restrict__ t_prime, const float* __restrict__ bias_p,
-    float* __restrict__ out,          // MODE 0: loss[1].  MODE 1: scal[2].
-    __bf16* __restrict__ g_out,       // MODE 1 only; leading dim = n
-    int b, int n, int d, int diag, int flags) {
-  __shared__ char smem[4 * TILE_BYTES];
-
+__device__ __forceinline__ void remap_block(int flags, int& bx, int& by) {
   const int gx = gridDim.x, gy = gridDim.y;
   int id = blockIdx.y * gx + blockIdx.x;
   if (flags & 1) {
-    // Bijective XCD-contiguous remap: each XCD gets a contiguous id span.
     const int nwg = gx * gy;
     const int q = nwg / NXCD, r = nwg % NXCD;
     const int xcd = id % NXCD, idx = id / NXCD;
     id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
-  int bx, by;
   if (flags & 2) {
-    // Grouped column-major walk: GROUP_M block-rows per group, so
-    // temporally-close blocks reuse both operand panels.
     const int group = id / (GROUP_M * gy);
     const int within = id % (GROUP_M * gy);
     const int gm = min(GROUP_M, gx - group * GROUP_M);
@@ -396,6 +366,37 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     bx = id % gx;
     by = id / gx;
   }
+}
+
+// Interior-only kernel: every tile full, d%64==0, n%8==0 — checked by the
+// host launcher.  Separate from the general kernel so the hot path's
+// register allocation is not inflated by the guarded path.
+template <int MODE>
+__launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
+    const __bf16* __restrict__ zimg, const __bf16* __restrict__ ztxt,
+    const float* __restrict__ t_prime, const float* __restrict__ bias_p,
+    float* __restrict__ out, __bf16* __restrict__ g_out,
+    int b, int n, int d, int diag, int flags) {
+  __shared__ char smem[4 * TILE_BYTES];
+  int bx, by;
+  remap_block(flags, bx, by);
+  const float t = __expf(*t_prime);
+  const float bias = *bias_p;
+  tile_body<MODE, true>(zimg, ztxt, t, bias, out, g_out, b, n, d, diag,
+                        bx * BM, by * BN, smem);
+}
+
+// MODE 0: forward loss.  MODE 1: backward g-slab + scalar partials.
+template <int MODE>
+__launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
+    const __bf16* __restrict__ zimg, const __bf16* __restrict__ ztxt,
+    const float* __restrict__ t_prime, const float* __restrict__ bias_p,
+    float* __restrict__ out,          // MODE 0: loss[1].  MODE 1: scal[2].
+    __bf16* __restrict__ g_out,       // MODE 1 only; leading dim = n
+    int b, int n, int d, int diag, int flags) {
+  __shared__ char smem[4 * TILE_BYTES];
+  int bx, by;
+  remap_block(flags, bx, by);
   const int row_base = bx * BM;
   const int col_base = by * BN;
 
@@ -420,18 +421,30 @@ extern "C" {
 
 int siglip_ext_abi(void) { return 3; }
 
+static inline bool all_interior(int b, int n, int d) {
+  return (b % BM == 0) && (n % BN == 0) && (d % BK == 0);
+}
+
 int siglip_fwd_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                     const void* t_prime, const void* bias, void* loss_out,
                     int b, int n, int d, int diag, int flags) {
   if (b <= 0 || n <= 0) return (int)hipSuccess;
   if (d % 8 != 0) return (int)hipErrorInvalidValue;
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
-  hipLaunchKernelGGL((siglip_tile_kernel<0>), grid, dim3(THREADS), 0,
-                     (hipStream_t)stream,
-                     (const __bf16*)zimg, (const __bf16*)ztxt,
-                     (const float*)t_prime, (const float*)bias,
-                     (float*)loss_out, (__bf16*)nullptr, b, n, d, diag,
-                     flags);
+  if (all_interior(b, n, d))
+    hipLaunchKernelGGL((siglip_tile_kernel_interior<0>), grid, dim3(THREADS),
+                       0, (hipStream_t)stream,
+                       (const __bf16*)zimg, (const __bf16*)ztxt,
+                       (const float*)t_prime, (const float*)bias,
+                       (float*)loss_out, (__bf16*)nullptr, b, n, d, diag,
+                       flags);
+  else
+    hipLaunchKernelGGL((siglip_tile_kernel<0>), grid, dim3(THREADS), 0,
+                       (hipStream_t)stream,
+                       (const __bf16*)zimg, (const __bf16*)ztxt,
+                       (const float*)t_prime, (const float*)bias,
+                       (float*)loss_out, (__bf16*)nullptr, b, n, d, diag,
+                       flags);
   return (int)hipGetLastError();
 }
 
@@ -441,11 +454,18 @@ int siglip_bwd_g_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
   if (b <= 0 || n <= 0) return (int)hipSuccess;
   if (d % 8 != 0) return (int)hipErrorInvalidValue;
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
-  hipLaunchKernelGGL((siglip_tile_kernel<1>), grid, dim3(THREADS), 0,
-                     (hipStream_t)stream,
-                     (const __bf16*)zimg, (const __bf16*)ztxt,
-                     (const float*)t_prime, (const float*)bias,
-                     (float*)scal, (__bf16*)g_out, b, n, d, diag, flags);
+  if (all_interior(b, n, d))
+    hipLaunchKernelGGL((siglip_tile_kernel_interior<1>), grid, dim3(THREADS),
+                       0, (hipStream_t)stream,
+                       (const __bf16*)zimg, (const __bf16*)ztxt,
+                       (const float*)t_prime, (const float*)bias,
+                       (float*)scal, (__bf16*)g_out, b, n, d, diag, flags);
+  else
+    hipLaunchKernelGGL((siglip_tile_kernel<1>), grid, dim3(THREADS), 0,
+                       (hipStream_t)stream,
+                       (const __bf16*)zimg, (const __bf16*)ztxt,
+                       (const float*)t_prime, (const float*)bias,
+                       (float*)scal, (__bf16*)g_out, b, n, d, diag, flags);
   return (int)hipGetLastError();
 }
 
