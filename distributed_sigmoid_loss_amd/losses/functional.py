@@ -131,8 +131,7 @@ def chunk_loss_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
     if impl == "hip":
         from .. import ops
         return ops.siglip_bwd(zimg.contiguous(), ztxt.contiguous(), t_prime,
-                              bias, diag_offset, grad_output,
-                              col_chunk or DEFAULT_COL_CHUNK)
+                              bias, diag_offset, grad_output, col_chunk)
     with torch.no_grad():
         return _torch_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
                           col_chunk)
@@ -149,7 +148,7 @@ class _FusedSigmoidLoss(torch.autograd.Function):
         loss = ops.siglip_fwd(zimg, ztxt, t_prime, bias, diag_offset)
         ctx.save_for_backward(zimg, ztxt, t_prime, bias)
         ctx.diag_offset = diag_offset
-        ctx.col_chunk = col_chunk or DEFAULT_COL_CHUNK
+        ctx.col_chunk = col_chunk   # None → single slab when addressable
         return loss
 
     @staticmethod

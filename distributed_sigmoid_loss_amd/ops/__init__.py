@@ -34,7 +34,7 @@ def _kernel_flags() -> int:
     # bit 1: grouped block walk for L2 panel reuse (default on;
     #        SIGLIP_GROUP_SWZ=0 disables for A/B profiling).
     f = 0
-    if os.environ.get("SIGLIP_XCD_SWZ", "0") == "1":
+    if os.environ.get("SIGLIP_XCD_SWZ", "1") != "0":
         f |= 1
     if os.environ.get("SIGLIP_GROUP_SWZ", "1") != "0":
         f |= 2
@@ -139,34 +139,53 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     scal = torch.zeros(2, device=dev, dtype=torch.float32)
     stream = torch.cuda.current_stream(dev).cuda_stream
 
-    dzimg_acc = torch.zeros((b, d), device=dev, dtype=torch.float32)
-    dztxt_acc = torch.empty((n, d), device=dev, dtype=torch.float32)
-
+    # Column slab sizing: the kernel's g-store addressing is 32-bit, so
+    # b * slab * 2 bytes must stay below 2^32; beyond that (and to bound
+    # workspace at huge n) we chunk.
     step = col_chunk if col_chunk and col_chunk > 0 else n
-    g_buf = torch.empty((b, min(step, n)), device=dev, dtype=torch.bfloat16)
-    for j0 in range(0, n, step):
-        j1 = min(j0 + step, n)
-        c = j1 - j0
-        zt = ztxt[j0:j1].contiguous()
-        g = g_buf[:, :c] if g_buf.shape[1] >= c else torch.empty(
-            (b, c), device=dev, dtype=torch.bfloat16)
-        if not g.is_contiguous():
-            g = torch.empty((b, c), device=dev, dtype=torch.bfloat16)
-        diag = _DIAG_NONE if diag_offset is None else int(diag_offset) - j0
-        _check(lib.siglip_bwd_g_bf16(
-            ctypes.c_void_p(stream),
-            ctypes.c_void_p(zimg.data_ptr()), ctypes.c_void_p(zt.data_ptr()),
-            ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
-            ctypes.c_void_p(g.data_ptr()), ctypes.c_void_p(scal.data_ptr()),
-            b, c, d, diag, _kernel_flags()), "siglip_bwd_g_bf16")
-        dzimg_acc += (g @ zt).float()
-        dztxt_acc[j0:j1] = (g.T @ zimg).float()
+    while (b * step * 2) >= 2 ** 32:
+        step //= 2
+    step = max(step, 256)
 
     t = tp.exp()
     go = grad_output.detach().reshape(()).to(device=dev, dtype=torch.float32)
     scale = go * t
-    dzimg = (dzimg_acc * scale).to(zimg.dtype)
-    dztxt = (dztxt_acc * scale).to(ztxt.dtype)
+
+    def run_g(zt_slab, g_slab, diag):
+        _check(lib.siglip_bwd_g_bf16(
+            ctypes.c_void_p(stream),
+            ctypes.c_void_p(zimg.data_ptr()),
+            ctypes.c_void_p(zt_slab.data_ptr()),
+            ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
+            ctypes.c_void_p(g_slab.data_ptr()),
+            ctypes.c_void_p(scal.data_ptr()),
+            b, zt_slab.shape[0], d, diag, _kernel_flags()),
+            "siglip_bwd_g_bf16")
+
+    if step >= n:
+        # Single-slab fast path: no fp32 accumulation round trips.
+        g = torch.empty((b, n), device=dev, dtype=torch.bfloat16)
+        run_g(ztxt, g, _DIAG_NONE if diag_offset is None else int(diag_offset))
+        dzimg = (g @ ztxt) * scale
+        dztxt = (g.T @ zimg) * scale
+    else:
+        dzimg_acc = torch.zeros((b, d), device=dev, dtype=torch.float32)
+        dztxt = torch.empty((n, d), device=dev, dtype=ztxt.dtype)
+        g_buf = torch.empty((b, step), device=dev, dtype=torch.bfloat16)
+        for j0 in range(0, n, step):
+            j1 = min(j0 + step, n)
+            c = j1 - j0
+            zt = ztxt[j0:j1]
+            g = g_buf if c == step else torch.empty(
+                (b, c), device=dev, dtype=torch.bfloat16)
+            diag = _DIAG_NONE if diag_offset is None else int(diag_offset) - j0
+            run_g(zt, g, diag)
+            dzimg_acc += (g @ zt).float()
+            dztxt[j0:j1] = (g.T @ zimg) * scale
+        dzimg = dzimg_acc * scale
+
+    dzimg = dzimg.to(zimg.dtype)
+    dztxt = dztxt.to(ztxt.dtype)
     dt_prime = (scal[0] * scale).to(t_prime.dtype).reshape(t_prime.shape)
     dbias = (scal[1] * go).to(bias.dtype).reshape(bias.shape)
     return dzimg, dztxt, dt_prime, dbias
