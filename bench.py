@@ -44,6 +44,8 @@ def parse_args():
     p.add_argument("--strategy", choices=["ring", "all_gather"],
                    default="ring")
     p.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
+    p.add_argument("--quant", choices=["bf16", "fp8"], default="bf16",
+                   help="logit-kernel compute dtype (fp8 = MX-scaled e4m3)")
     p.add_argument("--device", choices=["cuda", "cpu"], default=None)
     p.add_argument("--col-chunk", type=int, default=None,
                    help="column slab size for chunked negatives")
@@ -66,7 +68,8 @@ def main():
 
     model = TwoTowerModel(args.dim, args.dim).to(device=device, dtype=dtype)
     loss_mod = DistributedSigmoidLoss(b, strategy=args.strategy,
-                                      col_chunk=args.col_chunk).to(device)
+                                      col_chunk=args.col_chunk,
+                                      quant=args.quant).to(device)
     params = list(model.parameters()) + list(loss_mod.parameters())
     opt = torch.optim.SGD(params, lr=1e-4)
 
@@ -122,7 +125,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": args.dtype,
+            "dtype": args.dtype if args.quant == "bf16" else "fp8",
             "data": "synthetic",
             "config": {
                 "model": "two-tower-linear-siglip",

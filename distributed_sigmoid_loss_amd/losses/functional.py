@@ -110,7 +110,7 @@ def _torch_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
 
 
 def chunk_loss_fwd(zimg, ztxt, t_prime, bias, diag_offset=None,
-                   col_chunk=None, impl="auto"):
+                   col_chunk=None, impl="auto", quant="bf16"):
     """Non-differentiable forward of the block loss (used inside hand-written
     autograd Functions).  Returns a scalar tensor."""
     if impl == "auto":
@@ -118,20 +118,21 @@ def chunk_loss_fwd(zimg, ztxt, t_prime, bias, diag_offset=None,
     if impl == "hip":
         from .. import ops
         return ops.siglip_fwd(zimg.contiguous(), ztxt.contiguous(), t_prime,
-                              bias, diag_offset)
+                              bias, diag_offset, quant=quant)
     with torch.no_grad():
         return _torch_loss(zimg, ztxt, t_prime, bias, diag_offset, col_chunk)
 
 
 def chunk_loss_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
-                   col_chunk=None, impl="auto"):
+                   col_chunk=None, impl="auto", quant="bf16"):
     """Gradients of :func:`chunk_loss_fwd` wrt (zimg, ztxt, t_prime, bias)."""
     if impl == "auto":
         impl = "hip" if zimg.is_cuda else "torch"
     if impl == "hip":
         from .. import ops
         return ops.siglip_bwd(zimg.contiguous(), ztxt.contiguous(), t_prime,
-                              bias, diag_offset, grad_output, col_chunk)
+                              bias, diag_offset, grad_output, col_chunk,
+                              quant=quant)
     with torch.no_grad():
         return _torch_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
                           col_chunk)
@@ -141,14 +142,17 @@ class _FusedSigmoidLoss(torch.autograd.Function):
     """GPU path: hand-written HIP kernels; backward recomputes logits."""
 
     @staticmethod
-    def forward(ctx, zimg, ztxt, t_prime, bias, diag_offset, col_chunk):
+    def forward(ctx, zimg, ztxt, t_prime, bias, diag_offset, col_chunk,
+                quant):
         from .. import ops
         zimg = zimg.contiguous()
         ztxt = ztxt.contiguous()
-        loss = ops.siglip_fwd(zimg, ztxt, t_prime, bias, diag_offset)
+        loss = ops.siglip_fwd(zimg, ztxt, t_prime, bias, diag_offset,
+                              quant=quant)
         ctx.save_for_backward(zimg, ztxt, t_prime, bias)
         ctx.diag_offset = diag_offset
         ctx.col_chunk = col_chunk   # None → single slab when addressable
+        ctx.quant = quant
         return loss
 
     @staticmethod
@@ -157,15 +161,16 @@ class _FusedSigmoidLoss(torch.autograd.Function):
         zimg, ztxt, t_prime, bias = ctx.saved_tensors
         dzimg, dztxt, dt_prime, dbias = ops.siglip_bwd(
             zimg, ztxt, t_prime, bias, ctx.diag_offset, grad_output,
-            ctx.col_chunk)
-        return dzimg, dztxt, dt_prime, dbias, None, None
+            ctx.col_chunk, quant=ctx.quant)
+        return dzimg, dztxt, dt_prime, dbias, None, None, None
 
 
 def sigmoid_contrastive_loss(zimg: torch.Tensor, ztxt: torch.Tensor,
                              t_prime: torch.Tensor, bias: torch.Tensor,
                              diag_offset: Optional[int] = 0,
                              col_chunk: Optional[int] = None,
-                             impl: str = "auto") -> torch.Tensor:
+                             impl: str = "auto",
+                             quant: str = "bf16") -> torch.Tensor:
     """Sum of per-pair sigmoid cross-entropy over the ``(b, n)`` block.
 
     Args:
@@ -188,7 +193,7 @@ def sigmoid_contrastive_loss(zimg: torch.Tensor, ztxt: torch.Tensor,
         impl = "hip" if zimg.is_cuda else "torch"
     if impl == "hip":
         return _FusedSigmoidLoss.apply(zimg, ztxt, t_prime, bias, diag_offset,
-                                       col_chunk)
+                                       col_chunk, quant)
     if impl == "torch":
         return _torch_loss(zimg, ztxt, t_prime, bias, diag_offset, col_chunk)
     raise ValueError(f"unknown impl {impl!r}")

@@ -73,7 +73,8 @@ class _RingAllGatherLoss(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, zimg, ztxt, t_prime, bias, group, col_chunk, impl):
+    def forward(ctx, zimg, ztxt, t_prime, bias, group, col_chunk, impl,
+                quant="bf16"):
         world, rank = _world_and_rank(group)
         zimg = zimg.contiguous()
         ztxt = ztxt.contiguous()
@@ -89,8 +90,8 @@ class _RingAllGatherLoss(torch.autograd.Function):
             # local-block kernel below.
             handle = neighbour_exchange_start(left, right, ztxt, group=group)
 
-        loss = chunk_loss_fwd(zimg, ztxt, t_prime, bias,
-                              diag_offset=0, col_chunk=col_chunk, impl=impl)
+        loss = chunk_loss_fwd(zimg, ztxt, t_prime, bias, diag_offset=0,
+                              col_chunk=col_chunk, impl=impl, quant=quant)
 
         if world > 1:
             for hop in range(1, world):
@@ -103,7 +104,8 @@ class _RingAllGatherLoss(torch.autograd.Function):
                                                       group=group)
                 loss = loss + chunk_loss_fwd(zimg, recv, t_prime, bias,
                                              diag_offset=None,
-                                             col_chunk=col_chunk, impl=impl)
+                                             col_chunk=col_chunk, impl=impl,
+                                             quant=quant)
 
         ctx.save_for_backward(zimg, t_prime, bias, *chunks)
         ctx.group = group
@@ -111,6 +113,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
         ctx.rank = rank
         ctx.col_chunk = col_chunk
         ctx.impl = impl
+        ctx.quant = quant
         ctx.b_txt = b_txt
         return loss
 
@@ -127,7 +130,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
         all_txt = chunks[0] if world == 1 else torch.cat(chunks, dim=0)
         dzimg, dtxt_flat, dt_prime, dbias = chunk_loss_bwd(
             zimg, all_txt, t_prime, bias, rank * b, grad_output,
-            col_chunk=ctx.col_chunk, impl=ctx.impl)
+            col_chunk=ctx.col_chunk, impl=ctx.impl, quant=ctx.quant)
 
         if world > 1:
             # Each rank holds grad contributions for every rank's text shard;
@@ -146,7 +149,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
         else:
             dztxt = dtxt_flat
 
-        return dzimg, dztxt, dt_prime, dbias, None, None, None
+        return dzimg, dztxt, dt_prime, dbias, None, None, None, None
 
 
 class DistributedSigmoidLoss(nn.Module):
@@ -173,16 +176,20 @@ class DistributedSigmoidLoss(nn.Module):
     """
 
     def __init__(self, gpu_batch_size: int, strategy: str = "all_gather",
-                 col_chunk: Optional[int] = None, impl: str = "auto"):
+                 col_chunk: Optional[int] = None, impl: str = "auto",
+                 quant: str = "bf16"):
         super().__init__()
         self.t_prime = nn.Parameter(torch.tensor(math.log(10.0)))
         self.bias = nn.Parameter(torch.tensor(-10.0))
         self.gpu_batch_size = gpu_batch_size
         if strategy not in ("all_gather", "ring"):
             raise ValueError(f"unknown strategy {strategy!r}")
+        if quant not in ("bf16", "fp8"):
+            raise ValueError(f"unknown quant {quant!r}")
         self.strategy = strategy
         self.col_chunk = col_chunk
         self.impl = impl
+        self.quant = quant
 
     def forward(self, image_embeddings: torch.Tensor,
                 text_embeddings: torch.Tensor, group=None) -> torch.Tensor:
@@ -193,11 +200,11 @@ class DistributedSigmoidLoss(nn.Module):
             total = sigmoid_contrastive_loss(
                 image_embeddings, all_txt, self.t_prime, self.bias,
                 diag_offset=rank * b_txt, col_chunk=self.col_chunk,
-                impl=self.impl)
+                impl=self.impl, quant=self.quant)
         else:
             total = _RingAllGatherLoss.apply(
                 image_embeddings, text_embeddings, self.t_prime, self.bias,
-                group, self.col_chunk, self.impl)
+                group, self.col_chunk, self.impl, self.quant)
         return total / self.gpu_batch_size
 
 

@@ -63,6 +63,10 @@ def _load():
     lib.siglip_fwd_bf16.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] * 5
     lib.siglip_bwd_g_bf16.restype = ctypes.c_int
     lib.siglip_bwd_g_bf16.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] * 5
+    lib.siglip_fwd_fp8.restype = ctypes.c_int
+    lib.siglip_fwd_fp8.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] * 5
+    lib.siglip_bwd_g_fp8.restype = ctypes.c_int
+    lib.siglip_bwd_g_fp8.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] * 5
     _lib = lib
     return _lib
 
@@ -88,41 +92,70 @@ def _prep_scalar(p: torch.Tensor, device) -> torch.Tensor:
     return t.contiguous()
 
 
-def _validate(zimg: torch.Tensor, ztxt: torch.Tensor):
+def _validate(zimg: torch.Tensor, ztxt: torch.Tensor, quant: str):
     if not zimg.is_cuda:
         raise RuntimeError("siglip HIP ops require GPU tensors")
     if zimg.dtype != torch.bfloat16 or ztxt.dtype != torch.bfloat16:
         raise RuntimeError(
             f"siglip HIP ops require bf16 embeddings (got {zimg.dtype}); "
             "cast with .bfloat16() or pass impl='torch'")
-    if zimg.shape[1] % 8 != 0:
-        raise RuntimeError(f"emb dim must be a multiple of 8 (got {zimg.shape[1]})")
+    if quant not in ("bf16", "fp8"):
+        raise ValueError(f"unknown quant {quant!r}")
+    mult = 16 if quant == "fp8" else 8
+    if zimg.shape[1] % mult != 0:
+        raise RuntimeError(
+            f"emb dim must be a multiple of {mult} for {quant} "
+            f"(got {zimg.shape[1]})")
+
+
+def _quant_fp8(x: torch.Tensor):
+    """Per-tensor symmetric quantization to OCP e4m3 (max normal 448).
+
+    Returns (q, scale): x ≈ q * scale.  The scale is folded into the
+    temperature for the logit kernels (t_eff = t·s_img·s_txt), so the HIP
+    side runs with unit MX block scales.
+    """
+    amax = x.detach().abs().amax().float().clamp_(min=2.0 ** -20)
+    scale = amax / 448.0
+    q = (x.float() / scale).to(torch.float8_e4m3fn)
+    return q, scale
 
 
 def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
-               bias: torch.Tensor, diag_offset: Optional[int]) -> torch.Tensor:
+               bias: torch.Tensor, diag_offset: Optional[int],
+               quant: str = "bf16") -> torch.Tensor:
     lib = _require_lib()
-    _validate(zimg, ztxt)
+    _validate(zimg, ztxt, quant)
     b, d = zimg.shape
     n = ztxt.shape[0]
     dev = zimg.device
     tp = _prep_scalar(t_prime, dev)
     bp = _prep_scalar(bias, dev)
+    if quant == "fp8":
+        zi_q, si = _quant_fp8(zimg)
+        zt_q, st = _quant_fp8(ztxt)
+        tp = tp + si.log() + st.log()
+        zi_ptr, zt_ptr = zi_q.data_ptr(), zt_q.data_ptr()
+        fn = lib.siglip_fwd_fp8
+    else:
+        zi_ptr, zt_ptr = zimg.data_ptr(), ztxt.data_ptr()
+        fn = lib.siglip_fwd_bf16
     loss = torch.zeros((), device=dev, dtype=torch.float32)
     stream = torch.cuda.current_stream(dev).cuda_stream
     diag = _DIAG_NONE if diag_offset is None else int(diag_offset)
-    _check(lib.siglip_fwd_bf16(
+    _check(fn(
         ctypes.c_void_p(stream),
-        ctypes.c_void_p(zimg.data_ptr()), ctypes.c_void_p(ztxt.data_ptr()),
+        ctypes.c_void_p(zi_ptr), ctypes.c_void_p(zt_ptr),
         ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
         ctypes.c_void_p(loss.data_ptr()), b, n, d, diag, _kernel_flags()),
-        "siglip_fwd_bf16")
+        "siglip_fwd")
     return loss
 
 
 def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                bias: torch.Tensor, diag_offset: Optional[int],
-               grad_output: torch.Tensor, col_chunk: int):
+               grad_output: torch.Tensor, col_chunk: Optional[int],
+               quant: str = "bf16"):
     """Returns (dzimg, dztxt, dt_prime, dbias).
 
     Per column slab: the fused kernel recomputes logit tiles (MFMA) and writes
@@ -130,12 +163,25 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     dztxt_slab = gᵀ @ zimg (rocBLAS bf16 GEMMs, fp32 accumulation buffers).
     """
     lib = _require_lib()
-    _validate(zimg, ztxt)
+    _validate(zimg, ztxt, quant)
     b, d = zimg.shape
     n = ztxt.shape[0]
     dev = zimg.device
     tp = _prep_scalar(t_prime, dev)
     bp = _prep_scalar(bias, dev)
+    t_true = tp.exp()
+    if quant == "fp8":
+        # Same per-tensor scales as forward (identical inputs → identical
+        # amax); the g kernel sees t_eff so its logits match the forward's.
+        zi_q, si = _quant_fp8(zimg)
+        zt_q, st = _quant_fp8(ztxt)
+        tp_k = tp + si.log() + st.log()
+        g_fn = lib.siglip_bwd_g_fp8
+        zi_k, zt_k = zi_q, zt_q
+    else:
+        tp_k = tp
+        g_fn = lib.siglip_bwd_g_bf16
+        zi_k, zt_k = zimg, ztxt
     scal = torch.zeros(2, device=dev, dtype=torch.float32)
     stream = torch.cuda.current_stream(dev).cuda_stream
 
@@ -147,25 +193,25 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         step //= 2
     step = max(step, 256)
 
-    t = tp.exp()
     go = grad_output.detach().reshape(()).to(device=dev, dtype=torch.float32)
-    scale = go * t
+    scale = go * t_true   # gradient GEMMs run against the original bf16
 
-    def run_g(zt_slab, g_slab, diag):
-        _check(lib.siglip_bwd_g_bf16(
+    def run_g(j0, j1, g_slab, diag):
+        zt_slab = zt_k[j0:j1]
+        _check(g_fn(
             ctypes.c_void_p(stream),
-            ctypes.c_void_p(zimg.data_ptr()),
+            ctypes.c_void_p(zi_k.data_ptr()),
             ctypes.c_void_p(zt_slab.data_ptr()),
-            ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
+            ctypes.c_void_p(tp_k.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
             ctypes.c_void_p(g_slab.data_ptr()),
             ctypes.c_void_p(scal.data_ptr()),
-            b, zt_slab.shape[0], d, diag, _kernel_flags()),
-            "siglip_bwd_g_bf16")
+            b, j1 - j0, d, diag, _kernel_flags()),
+            "siglip_bwd_g")
 
     if step >= n:
         # Single-slab fast path: no fp32 accumulation round trips.
         g = torch.empty((b, n), device=dev, dtype=torch.bfloat16)
-        run_g(ztxt, g, _DIAG_NONE if diag_offset is None else int(diag_offset))
+        run_g(0, n, g, _DIAG_NONE if diag_offset is None else int(diag_offset))
         dzimg = (g @ ztxt) * scale
         dztxt = (g.T @ zimg) * scale
     else:
@@ -175,17 +221,18 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         for j0 in range(0, n, step):
             j1 = min(j0 + step, n)
             c = j1 - j0
-            zt = ztxt[j0:j1]
             g = g_buf if c == step else torch.empty(
                 (b, c), device=dev, dtype=torch.bfloat16)
             diag = _DIAG_NONE if diag_offset is None else int(diag_offset) - j0
-            run_g(zt, g, diag)
-            dzimg_acc += (g @ zt).float()
+            run_g(j0, j1, g, diag)
+            dzimg_acc += (g @ ztxt[j0:j1]).float()
             dztxt[j0:j1] = (g.T @ zimg) * scale
         dzimg = dzimg_acc * scale
 
     dzimg = dzimg.to(zimg.dtype)
     dztxt = dztxt.to(ztxt.dtype)
-    dt_prime = (scal[0] * scale).to(t_prime.dtype).reshape(t_prime.shape)
+    # dt' = go · Σ g·(z−bias) = go · t_eff · Σ g·dot_q  (t_eff ≡ t for bf16).
+    t_eff = tp_k.exp()
+    dt_prime = (scal[0] * go * t_eff).to(t_prime.dtype).reshape(t_prime.shape)
     dbias = (scal[1] * go).to(bias.dtype).reshape(bias.shape)
     return dzimg, dztxt, dt_prime, dbias
