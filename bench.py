@@ -46,6 +46,8 @@ def parse_args():
     p.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
     p.add_argument("--quant", choices=["bf16", "fp8"], default="bf16",
                    help="logit-kernel compute dtype (fp8 = MX-scaled e4m3)")
+    p.add_argument("--impl", choices=["auto", "hip", "torch"], default="auto",
+                   help="'torch' = stock-PyTorch floor (materialized logits)")
     p.add_argument("--device", choices=["cuda", "cpu"], default=None)
     p.add_argument("--col-chunk", type=int, default=None,
                    help="column slab size for chunked negatives")
@@ -69,7 +71,8 @@ def main():
     model = TwoTowerModel(args.dim, args.dim).to(device=device, dtype=dtype)
     loss_mod = DistributedSigmoidLoss(b, strategy=args.strategy,
                                       col_chunk=args.col_chunk,
-                                      quant=args.quant).to(device)
+                                      quant=args.quant,
+                                      impl=args.impl).to(device)
     params = list(model.parameters()) + list(loss_mod.parameters())
     opt = torch.optim.SGD(params, lr=1e-4)
 
@@ -133,6 +136,7 @@ def main():
                 "emb_dim": args.dim,
                 "seq_len": None,
                 "parallelism": f"dp{world}-{args.strategy}",
+                "impl": args.impl,
             },
         }), flush=True)
 

@@ -19,10 +19,22 @@ double-buffered schedule (the reference serializes comm and compute by calling
 
 from __future__ import annotations
 
+import os
 from typing import Optional, Sequence
 
 import torch
 import torch.distributed as dist
+
+# Debug escape hatch (SURVEY §5 race-detection plan): SIGLIP_SYNC_COMM=1
+# makes every exchange complete before returning, serializing comm against
+# compute — use to bisect overlap bugs (with HIP_LAUNCH_BLOCKING=1 for
+# kernel-side ordering).
+_SYNC_COMM = os.environ.get("SIGLIP_SYNC_COMM", "0") == "1"
+
+
+def set_sync_comm(enabled: bool) -> None:
+    global _SYNC_COMM
+    _SYNC_COMM = enabled
 
 
 class RingHandle:
@@ -59,7 +71,10 @@ def isend_irecv(sends, send_ranks, recvs, recv_ranks, group=None) -> RingHandle:
     for t, r in zip(recvs, recv_ranks):
         ops.append(dist.P2POp(dist.irecv, t, r, group=group))
     reqs = dist.batch_isend_irecv(ops)
-    return RingHandle(reqs, list(recvs))
+    handle = RingHandle(reqs, list(recvs))
+    if _SYNC_COMM:
+        handle.wait()
+    return handle
 
 
 def neighbour_exchange(from_rank: int, to_rank: int, tensor: torch.Tensor,
