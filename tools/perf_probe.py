@@ -58,6 +58,15 @@ def main():
     print(f"fwd fused kernel      : {t_fwd:8.3f} ms   "
           f"{flops_fwd / t_fwd / 1e9:7.1f} TF/s")
 
+    t_fwd8 = time_fn(lambda: ops.siglip_fwd(zi, zt, tp, bs, 0, quant="fp8"))
+    print(f"fwd fused fp8 (MX)    : {t_fwd8:8.3f} ms   "
+          f"{flops_fwd / t_fwd8 / 1e9:7.1f} TF/s  (incl quantize)")
+
+    go1 = torch.tensor(1.0, device=dev)
+    t_bwd8 = time_fn(lambda: ops.siglip_bwd(zi, zt, tp, bs, 0, go1, None,
+                                            quant="fp8"))
+    print(f"bwd total fp8 (MX)    : {t_bwd8:8.3f} ms")
+
     # rocBLAS ceiling reference: the same-shape plain GEMM (materializes the
     # logits we fuse away) — an upper bound on achievable MFMA throughput.
     ref_n = min(b, 8192)
