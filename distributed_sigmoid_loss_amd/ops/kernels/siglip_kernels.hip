@@ -298,11 +298,54 @@ __device__ __forceinline__ void tile_body(
         }
       }
     }
+  } else if (INTERIOR) {
+    // Stage this wave's 128×64 g sub-tile in its own LDS slice (the staging
+    // buffers are free after the main loop; no cross-wave barrier needed),
+    // then store it as 16 rounds of 8 fully-coalesced 128-B rows — 16
+    // dwordx4 stores per lane instead of 128 scattered 2-B stores.
+    char* wlds = smem + wave * (128 * 128);   // 16 KiB per wave
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int lr = mi * 16 + (lane >> 4) * 4 + reg;   // 0..127
+          const int ec = ni * 16 + (lane & 15);             // 0..63
+          const int grow = row_base + wrow + lr;
+          const int gcol = col_base + wcol + ec;
+          const float dot = acc[mi][ni][reg];
+          const float z = dot * t + bias;
+          const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
+          const float gv = sigmoid_fast(pos ? z : -z);
+          const float g = pos ? -gv : gv;
+          // Swizzled 16-B chunk index so the readback is conflict-free.
+          *reinterpret_cast<__bf16*>(
+              wlds + lr * 128 + (((ec >> 3) ^ kmask(lr)) * 16)
+              + (ec & 7) * 2) = (__bf16)g;
+          s0 += g * dot;
+          s1 += g;
+        }
+      }
+    }
+    // Own-wave LDS round trip: the compiler's lgkmcnt tracking orders the
+    // reads after the writes; other waves never touch this slice.
+    {
+      const int rsub8 = lane >> 3;     // 8 rows per round
+      const int c8 = lane & 7;         // 16-B chunk within the row
+      __bf16* gb2 = g_out + (size_t)(row_base + wrow) * n + col_base + wcol;
+#pragma unroll
+      for (int r0 = 0; r0 < 128; r0 += 8) {
+        const int lr = r0 + rsub8;
+        const uint4 v = *reinterpret_cast<const uint4*>(
+            wlds + lr * 128 + ((c8 ^ kmask(lr)) * 16));
+        *reinterpret_cast<uint4*>(
+            reinterpret_cast<char*>(gb2) + (size_t)lr * n * 2 + c8 * 16) = v;
+      }
+    }
   } else {
     // One per-lane base offset + per-(mi,reg) scalar row offset keeps the
-    // store addressing affine — per-element (size_t)grow*n math made the
-    // allocator hoist 128 addresses and spill.  Caller guarantees
-    // b*n*2 < 2^32 (ops/__init__.py column-chunks the slab).
+    // store addressing affine.  (Edge blocks only.)
     __bf16* gb = g_out + (size_t)row_base * n + col_base;
     const unsigned lane_off =
         (unsigned)(wrow + (lane >> 4) * 4) * (unsigned)n
@@ -316,7 +359,7 @@ __device__ __forceinline__ void tile_body(
 #pragma unroll
         for (int ni = 0; ni < FN; ++ni) {
           const int gcol = col_base + wcol + ni * 16 + (lane & 15);
-          if (INTERIOR || (grow < b && gcol < n)) {
+          if (grow < b && gcol < n) {
             const float dot = acc[mi][ni][reg];
             const float z = dot * t + bias;
             const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
