@@ -48,6 +48,7 @@
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 #include <climits>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
@@ -386,6 +387,164 @@ __device__ __forceinline__ void tile_body(
   }
 }
 
+// ---------------- v10: 256² bf16 tile, 3-buffer depth-2 DMA ----------------
+// Same 8-wave 256×256 tile, but the K-step drops to 64 B rows (32 bf16) so
+// THREE K-step buffers fit LDS (3 × 32 KiB = 96 KiB): the DMA for tile k+2
+// is issued while tile k computes, giving each transfer a two-compute-phase
+// window instead of one — the measured per-iteration vmcnt wait disappears.
+// bf16 interior only; fp8 and ragged shapes use the kernels above.
+constexpr int VROW = 64;                     // bytes per LDS row (32 bf16)
+constexpr int VTILE = BM * VROW;             // 16 KiB per operand tile
+constexpr int VBUF = 2 * VTILE;              // 32 KiB per K-step buffer
+
+// 2-bit chunk mask for 64-B rows: rows sharing a bank base (r ≡ r' mod 4)
+// get distinct chunks; interleaved groups are conflict-free by row spacing.
+__device__ __forceinline__ int kmask2(int r) { return (r >> 2) & 3; }
+
+template <int MODE>
+__device__ __forceinline__ void tile_body_v10(
+    const char* __restrict__ zimg, const char* __restrict__ ztxt,
+    float t, float bias, float* __restrict__ out, __bf16* __restrict__ g_out,
+    int b, int n, int d, int diag, int row_base, int col_base, char* smem) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wrow = (wave >> 2) * 128;
+  const int wcol = (wave & 3) * 64;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < FN; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int ktiles = d / 32;                  // launcher enforces d%64==0
+  const int fr = lane & 15;
+  const int q = lane >> 4;                    // 16-B chunk 0..3
+
+  int aAddr = (wrow + fr) * VROW + ((q ^ kmask2(wrow + fr)) * 16);
+  int bAddr = VTILE + (wcol + fr) * VROW + ((q ^ kmask2(wcol + fr)) * 16);
+
+  // DMA: 16 KiB per operand per K-step = 16 wave-instructions; wave w
+  // issues 2 per operand (rows (w*2+j)*16 .. +16; 4 lanes per 64-B row).
+  const int rsub = lane >> 2;
+  const int cch = lane & 3;
+  int va[2];
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    const int rloc = (wave * 2 + j) * 16 + rsub;
+    va[j] = rloc * d * 2 + ((cch ^ kmask2(rloc)) * 16);
+  }
+  const char* abase = zimg + (size_t)row_base * d * 2;
+  const char* bbase = ztxt + (size_t)col_base * d * 2;
+
+  auto stage = [&](int buf) {
+    const int lb = __builtin_amdgcn_readfirstlane(wave * 2048) + buf * VBUF;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      __builtin_amdgcn_global_load_lds(
+          (gas_ptr)(abase + va[j]), (las_ptr)(smem + lb + j * 1024),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (gas_ptr)(bbase + va[j]),
+          (las_ptr)(smem + lb + VTILE + j * 1024), 16, 0, 0);
+    }
+    abase += VROW;
+    bbase += VROW;
+  };
+
+  stage(0);
+  if (ktiles > 1) stage(1);
+  int bufoff = 0;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    if (kt + 2 < ktiles) {
+      // Buffer (kt+2)%3 held tile kt-1; its reads were fenced by the
+      // barrier that ended iteration kt-1.
+      stage((kt + 2) % 3);
+      asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+    } else if (kt + 1 < ktiles) {
+      asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
+    }
+    {
+      const int aCur = aAddr + bufoff;
+      const int bCur = bAddr + bufoff;
+      bf16x8 afrag[FM], bfrag[FN];
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+        afrag[mi] = *reinterpret_cast<const bf16x8*>(
+            smem + aCur + mi * (16 * VROW));
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni)
+        bfrag[ni] = *reinterpret_cast<const bf16x8*>(
+            smem + bCur + ni * (16 * VROW));
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    bufoff = (bufoff == 2 * VBUF) ? 0 : bufoff + VBUF;
+    asm volatile("s_barrier" ::: "memory");
+  }
+
+  // Epilogue (identical math to the 256² kernels above).
+  float s0 = 0.f, s1 = 0.f;
+  if (MODE == 0) {
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int grow = row_base + wrow + mi * 16 + (lane >> 4) * 4 + reg;
+          const int gcol = col_base + wcol + ni * 16 + (lane & 15);
+          const float z = acc[mi][ni][reg] * t + bias;
+          const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
+          s0 += softplus_f(pos ? -z : z);
+        }
+      }
+    }
+  } else {
+    __bf16* gb = g_out + (size_t)row_base * n + col_base;
+    const unsigned lane_off =
+        (unsigned)(wrow + (lane >> 4) * 4) * (unsigned)n
+        + (unsigned)(wcol + (lane & 15));
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)n;
+        const int grow = row_base + wrow + mi * 16 + (lane >> 4) * 4 + reg;
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni) {
+          const int gcol = col_base + wcol + ni * 16 + (lane & 15);
+          const float dot = acc[mi][ni][reg];
+          const float z = dot * t + bias;
+          const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
+          const float gv = sigmoid_fast(pos ? z : -z);
+          const float g = pos ? -gv : gv;
+          gb[lane_off + row_off + ni * 16] = (__bf16)g;
+          s0 += g * dot;
+          s1 += g;
+        }
+      }
+      __builtin_amdgcn_sched_barrier(0);
+    }
+  }
+
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s0 += __shfl_down(s0, off);
+    if (MODE == 1) s1 += __shfl_down(s1, off);
+  }
+  if (lane == 0) {
+    atomicAdd(&out[0], s0);
+    if (MODE == 1) atomicAdd(&out[1], s1);
+  }
+}
+
 __device__ __forceinline__ void remap_block(int flags, int& bx, int& by) {
   const int gx = gridDim.x, gy = gridDim.y;
   int id = blockIdx.y * gx + blockIdx.x;
@@ -405,6 +564,21 @@ __device__ __forceinline__ void remap_block(int flags, int& bx, int& by) {
     bx = id % gx;
     by = id / gx;
   }
+}
+
+template <int MODE>
+__launch_bounds__(THREADS) __global__ void siglip_tile_kernel_v10(
+    const char* __restrict__ zimg, const char* __restrict__ ztxt,
+    const float* __restrict__ t_prime, const float* __restrict__ bias_p,
+    float* __restrict__ out, __bf16* __restrict__ g_out,
+    int b, int n, int d, int diag, int flags) {
+  __shared__ char smem[3 * VBUF];
+  int bx, by;
+  remap_block(flags, bx, by);
+  const float t = __expf(*t_prime);
+  const float bias = *bias_p;
+  tile_body_v10<MODE>(zimg, ztxt, t, bias, out, g_out, b, n, d, diag,
+                      bx * BM, by * BN, smem);
 }
 
 // Interior-only kernel: every tile full, d % K-step == 0, n%8==0 — checked
@@ -455,6 +629,15 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
 
 inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
+inline bool siglip_use_v10() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("SIGLIP_V10");
+    v = (e == nullptr || e[0] != '0') ? 1 : 0;   // default on; =0 for A/B
+  }
+  return v == 1;
+}
+
 template <int MODE, int EB>
 int launch(uintptr_t stream, const void* zimg, const void* ztxt,
            const void* t_prime, const void* bias, void* out, void* g_out,
@@ -462,6 +645,15 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
   if (b <= 0 || n <= 0) return (int)hipSuccess;
   if (d % (16 / EB) != 0) return (int)hipErrorInvalidValue;
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
+  if (EB == 2 && (b % BM == 0) && (n % BN == 0) && (d % 64 == 0) &&
+      siglip_use_v10()) {
+    hipLaunchKernelGGL((siglip_tile_kernel_v10<MODE>), grid, dim3(THREADS),
+                       0, (hipStream_t)stream,
+                       (const char*)zimg, (const char*)ztxt,
+                       (const float*)t_prime, (const float*)bias,
+                       (float*)out, (__bf16*)g_out, b, n, d, diag, flags);
+    return (int)hipGetLastError();
+  }
   const bool interior = (b % BM == 0) && (n % BN == 0) &&
       (d % (ROW_BYTES / EB) == 0);
   if (interior)
