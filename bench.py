@@ -31,6 +31,7 @@ from distributed_sigmoid_loss_amd import DistributedSigmoidLoss
 from distributed_sigmoid_loss_amd.models import TwoTowerModel
 from distributed_sigmoid_loss_amd.parallel import average_gradients
 from distributed_sigmoid_loss_amd.utils import init_from_env, set_seed
+from distributed_sigmoid_loss_amd.utils.profiling import PhaseTimer
 
 
 def parse_args():
@@ -51,6 +52,12 @@ def parse_args():
     p.add_argument("--device", choices=["cuda", "cpu"], default=None)
     p.add_argument("--col-chunk", type=int, default=None,
                    help="column slab size for chunked negatives")
+    p.add_argument("--csv", default=None,
+                   help="write per-step phase timings (ms) to this CSV")
+    p.add_argument("--ddp", action="store_true",
+                   help="wrap the towers in torch DDP (bucketed RCCL "
+                        "all-reduce overlapped with backward) instead of "
+                        "manual grad averaging")
     return p.parse_args()
 
 
@@ -73,21 +80,34 @@ def main():
                                       col_chunk=args.col_chunk,
                                       quant=args.quant,
                                       impl=args.impl).to(device)
+    if args.ddp and world > 1:
+        model = torch.nn.parallel.DistributedDataParallel(
+            model, device_ids=[local_rank] if device == "cuda" else None)
+        # loss params still averaged manually (2 scalars)
     params = list(model.parameters()) + list(loss_mod.parameters())
     opt = torch.optim.SGD(params, lr=1e-4)
+    timer = PhaseTimer(enabled=args.csv is not None,
+                       use_cuda=(device == "cuda"))
 
     img_feats = torch.randn(b, args.dim, device=device, dtype=dtype)
     txt_feats = torch.randn(b, args.dim, device=device, dtype=dtype)
 
     def step():
         opt.zero_grad(set_to_none=True)
-        zi, zt = model(img_feats, txt_feats)
-        loss = loss_mod(zi, zt)
-        loss.backward()
-        if world > 1:
-            average_gradients(model)
-            average_gradients(loss_mod)
-        opt.step()
+        with timer.phase("encode"):
+            zi, zt = model(img_feats, txt_feats)
+        with timer.phase("loss_fwd"):
+            loss = loss_mod(zi, zt)
+        with timer.phase("backward"):
+            loss.backward()
+        with timer.phase("grad_avg"):
+            if world > 1:
+                if not args.ddp:
+                    average_gradients(model)
+                average_gradients(loss_mod)
+        with timer.phase("optimizer"):
+            opt.step()
+        timer.step_end()
         return loss
 
     def sync():
@@ -112,6 +132,9 @@ def main():
             t = t.to(device)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
+
+    if args.csv and rank == 0:
+        timer.write_csv(args.csv)
 
     ms_per_step = elapsed / args.steps * 1000.0
     pairs_per_sec = args.global_batch / (elapsed / args.steps)

@@ -26,3 +26,42 @@ def test_bench_cpu_plumbing():
     assert out["n_gpus"] == 1
     assert out["value"] > 0
     assert out["data"] == "synthetic"
+
+
+def test_bench_distributed_cpu_2proc():
+    """The exact launch shape the driver uses (torch.distributed.run,
+    nnodes=1, master 127.0.0.1), world_size 2 on CPU/gloo."""
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--device", "cpu", "--global-batch", "8", "--dim",
+         "64", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert res.returncode == 0, res.stderr[-2000:]
+    lines = [l for l in res.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, res.stdout
+    out = json.loads(lines[0])
+    assert out["n_gpus"] == 2
+
+
+def test_bench_distributed_cpu_2proc_allgather_ddp():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--device", "cpu", "--global-batch", "8", "--dim", "64",
+         "--steps", "2", "--warmup", "1", "--strategy", "all_gather",
+         "--ddp"],
+        capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert res.returncode == 0, res.stderr[-2000:]
