@@ -124,18 +124,27 @@ def chunk_loss_fwd(zimg, ztxt, t_prime, bias, diag_offset=None,
 
 
 def chunk_loss_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
-                   col_chunk=None, impl="auto", quant="bf16"):
-    """Gradients of :func:`chunk_loss_fwd` wrt (zimg, ztxt, t_prime, bias)."""
+                   col_chunk=None, impl="auto", quant="bf16",
+                   on_dztxt=None):
+    """Gradients of :func:`chunk_loss_fwd` wrt (zimg, ztxt, t_prime, bias).
+
+    ``on_dztxt(dztxt)`` (optional) is invoked as soon as the text gradient is
+    materialized, before the image-gradient GEMM — lets a distributed caller
+    start its reduce-scatter while the remaining GEMM runs.
+    """
     if impl == "auto":
         impl = "hip" if zimg.is_cuda else "torch"
     if impl == "hip":
         from .. import ops
         return ops.siglip_bwd(zimg.contiguous(), ztxt.contiguous(), t_prime,
                               bias, diag_offset, grad_output, col_chunk,
-                              quant=quant)
+                              quant=quant, on_dztxt=on_dztxt)
     with torch.no_grad():
-        return _torch_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
-                          col_chunk)
+        out = _torch_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
+                         col_chunk)
+        if on_dztxt is not None:
+            on_dztxt(out[1])
+        return out
 
 
 class _FusedSigmoidLoss(torch.autograd.Function):

@@ -34,7 +34,6 @@ from .functional import (
     sigmoid_contrastive_loss,
     chunk_loss_fwd,
     chunk_loss_bwd,
-    DEFAULT_COL_CHUNK,
 )
 from ..parallel.collectives import all_gather_with_grad, _backend_is_gloo
 from ..parallel.ring import (
@@ -126,26 +125,41 @@ class _RingAllGatherLoss(torch.autograd.Function):
 
         # One fused backward over the concatenated (W·b, d) text block —
         # identical math to per-chunk calls (diag at the own-rank block),
-        # but a single dzimg accumulator and one slab loop.
+        # but a single dzimg accumulator and one slab loop.  The cross-rank
+        # reduce-scatter (the collapse of the reference's W−1 reversed ring
+        # hops, distributed_utils.py:74-77, into one RCCL collective) is
+        # launched ASYNC from the on_dztxt hook, so its xGMI wire time
+        # overlaps the dzimg gradient GEMM still running on the compute
+        # stream.
         all_txt = chunks[0] if world == 1 else torch.cat(chunks, dim=0)
+        comm = {}
+
+        def on_dztxt(flat):
+            if world == 1:
+                return
+            if _backend_is_gloo(ctx.group):
+                comm["work"] = dist.all_reduce(
+                    flat, op=dist.ReduceOp.SUM, group=ctx.group,
+                    async_op=True)
+                comm["flat"] = flat
+            else:
+                out = torch.empty_like(flat[:b])
+                comm["work"] = dist.reduce_scatter_tensor(
+                    out, flat, op=dist.ReduceOp.SUM, group=ctx.group,
+                    async_op=True)
+                comm["out"] = out
+
         dzimg, dtxt_flat, dt_prime, dbias = chunk_loss_bwd(
             zimg, all_txt, t_prime, bias, rank * b, grad_output,
-            col_chunk=ctx.col_chunk, impl=ctx.impl, quant=ctx.quant)
+            col_chunk=ctx.col_chunk, impl=ctx.impl, quant=ctx.quant,
+            on_dztxt=on_dztxt)
 
         if world > 1:
-            # Each rank holds grad contributions for every rank's text shard;
-            # one reduce-scatter(SUM) delivers each shard's total home — the
-            # collapse of the reference's W−1 reversed ring hops
-            # (distributed_utils.py:74-77) into a single RCCL collective.
-            if _backend_is_gloo(ctx.group):
-                dist.all_reduce(dtxt_flat, op=dist.ReduceOp.SUM,
-                                group=ctx.group)
-                dztxt = dtxt_flat[rank * b:(rank + 1) * b].clone()
+            comm["work"].wait()
+            if "out" in comm:
+                dztxt = comm["out"]
             else:
-                dztxt = torch.empty_like(dtxt_flat[:b])
-                dist.reduce_scatter_tensor(dztxt, dtxt_flat,
-                                           op=dist.ReduceOp.SUM,
-                                           group=ctx.group)
+                dztxt = comm["flat"][rank * b:(rank + 1) * b].clone()
         else:
             dztxt = dtxt_flat
 

@@ -155,7 +155,7 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
 def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                bias: torch.Tensor, diag_offset: Optional[int],
                grad_output: torch.Tensor, col_chunk: Optional[int],
-               quant: str = "bf16"):
+               quant: str = "bf16", on_dztxt=None):
     """Returns (dzimg, dztxt, dt_prime, dbias).
 
     Per column slab: the fused kernel recomputes logit tiles (MFMA) and writes
@@ -209,11 +209,15 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
             "siglip_bwd_g")
 
     if step >= n:
-        # Single-slab fast path: no fp32 accumulation round trips.
+        # Single-slab fast path: no fp32 accumulation round trips.  dztxt is
+        # produced FIRST so the caller's on_dztxt hook (e.g. an async RCCL
+        # reduce-scatter) overlaps with the dzimg GEMM below.
         g = torch.empty((b, n), device=dev, dtype=torch.bfloat16)
         run_g(0, n, g, _DIAG_NONE if diag_offset is None else int(diag_offset))
+        dztxt = ((g.T @ zimg) * scale).to(ztxt.dtype)
+        if on_dztxt is not None:
+            on_dztxt(dztxt)
         dzimg = (g @ ztxt) * scale
-        dztxt = (g.T @ zimg) * scale
     else:
         dzimg_acc = torch.zeros((b, d), device=dev, dtype=torch.float32)
         dztxt = torch.empty((n, d), device=dev, dtype=ztxt.dtype)
@@ -227,6 +231,8 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
             run_g(j0, j1, g, diag)
             dzimg_acc += (g @ ztxt[j0:j1]).float()
             dztxt[j0:j1] = (g.T @ zimg) * scale
+        if on_dztxt is not None:
+            on_dztxt(dztxt)
         dzimg = dzimg_acc * scale
 
     dzimg = dzimg.to(zimg.dtype)

@@ -173,3 +173,39 @@ def test_extension_is_intree_so():
     ops._require_lib()
     with open("/proc/self/maps") as f:
         assert any("_siglip_hip.so" in line for line in f)
+
+
+def test_siglip_loss_module_gpu():
+    """SigLipLoss (caller-owned params, reference rwightman API) through the
+    fused kernels on one GPU vs CPU fp32."""
+    from distributed_sigmoid_loss_amd import SigLipLoss
+    b, d = 256, 768
+    g = torch.Generator().manual_seed(12)
+    zi32 = F.normalize(torch.randn(b, d, generator=g), dim=-1)
+    zt32 = F.normalize(torch.randn(b, d, generator=g), dim=-1)
+    scale32 = torch.tensor(math.log(10.0))
+    bias32 = torch.tensor(-10.0)
+
+    mod = SigLipLoss(rank=0, world_size=1)
+    zi = zi32.cuda().bfloat16().requires_grad_(True)
+    zt = zt32.cuda().bfloat16().requires_grad_(True)
+    sc = scale32.cuda().requires_grad_(True)
+    bs = bias32.cuda().requires_grad_(True)
+    loss = mod(zi, zt, sc, bs)
+    loss.backward()
+
+    zi_c = zi32.clone().requires_grad_(True)
+    zt_c = zt32.clone().requires_grad_(True)
+    sc_c = scale32.clone().requires_grad_(True)
+    bs_c = bias32.clone().requires_grad_(True)
+    loss_c = mod(zi_c, zt_c, sc_c, bs_c)
+    loss_c.backward()
+    torch.cuda.synchronize()
+
+    assert torch.allclose(loss.cpu().float(), loss_c, rtol=2e-2, atol=1e-2)
+    assert torch.allclose(zi.grad.cpu().float(), zi_c.grad, rtol=5e-2,
+                          atol=5e-4)
+    assert torch.allclose(sc.grad.cpu().float(), sc_c.grad, rtol=2e-2,
+                          atol=1e-3)
+    assert torch.allclose(bs.grad.cpu().float(), bs_c.grad, rtol=2e-2,
+                          atol=1e-3)
