@@ -6,6 +6,11 @@
 //   2 = no staging/waits             (ds_read + MFMA on stale LDS)
 //   3 = MFMA only                    (fragments kept live via asm)
 //   4 = no MFMA                      (stage + waits + ds_read only)
+//   5 = decoupled DMA                (stage into unread dummy region; compute
+//                                     on static buffers — co-run without the
+//                                     producer-consumer coupling)
+//   6 = register DMA                 (plain global_load to discarded regs; no
+//                                     LDS writes — HBM read + MFMA co-run)
 // Results are wrong for ABL>0 — perf-diagnostic only.  Prints ms and TF/s
 // per variant, interleaved rounds (guide §5.4 rule 24).
 //
@@ -86,9 +91,43 @@ __launch_bounds__(THREADS) __global__ void fwd_kernel(
     bbase += 128;
   };
 
-  if (ABL != 2 && ABL != 3) stage(0);
+  auto stage_dummy = [&]() {
+    // DMA the same bytes into buffer 1's region, never read by compute.
+    const int lb = __builtin_amdgcn_readfirstlane(wave * 4096) +
+        2 * TILE_BYTES;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      __builtin_amdgcn_global_load_lds((gas_ptr)(abase + va[j]),
+                                       (las_ptr)(smem + lb + j * 1024), 16,
+                                       0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (gas_ptr)(bbase + va[j]), (las_ptr)(smem + lb + TILE_BYTES + j * 1024),
+          16, 0, 0);
+    }
+    abase += 128;
+    bbase += 128;
+  };
+  auto load_regs = [&]() {
+    // Same HBM traffic, straight to registers, discarded.
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      uint4 va_ = *reinterpret_cast<const uint4*>(abase + va[j]);
+      uint4 vb_ = *reinterpret_cast<const uint4*>(bbase + va[j]);
+      asm volatile("" :: "v"(va_), "v"(vb_));
+    }
+    abase += 128;
+    bbase += 128;
+  };
+
+  if (ABL != 2 && ABL != 3 && ABL != 5 && ABL != 6) stage(0);
   for (int kt = 0; kt < ktiles; ++kt) {
-    if (ABL != 2 && ABL != 3) {
+    if (ABL == 5) {
+      stage_dummy();
+      asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+    } else if (ABL == 6) {
+      load_regs();
+      asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+    } else if (ABL != 2 && ABL != 3) {
       if (kt + 1 < ktiles) {
         stage((kt + 1) & 1);
         asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
@@ -137,7 +176,8 @@ __launch_bounds__(THREADS) __global__ void fwd_kernel(
     }
     aAddr[0] ^= 2 * TILE_BYTES; aAddr[1] ^= 2 * TILE_BYTES;
     bAddr[0] ^= 2 * TILE_BYTES; bAddr[1] ^= 2 * TILE_BYTES;
-    if (ABL == 0) asm volatile("s_barrier" ::: "memory");
+    if (ABL == 0 || ABL == 5 || ABL == 6)
+      asm volatile("s_barrier" ::: "memory");
     else if (ABL == 2 || ABL == 3 || ABL == 4) __syncthreads();
     // ABL 1: no post-compute barrier
   }
@@ -201,16 +241,18 @@ int main() {
 
   const double flops = 2.0 * b * n * d;
   const char* names[] = {"full", "no-2nd-barrier", "no-stage", "mfma-only",
-                         "no-mfma"};
-  float best[5] = {1e9f, 1e9f, 1e9f, 1e9f, 1e9f};
+                         "no-mfma", "decoupled-dma", "reg-dma"};
+  float best[7] = {1e9f, 1e9f, 1e9f, 1e9f, 1e9f, 1e9f, 1e9f};
   for (int r = 0; r < rounds; ++r) {
     best[0] = fminf(best[0], run<0>(zi, zt, out, b, n, d, iters));
     best[1] = fminf(best[1], run<1>(zi, zt, out, b, n, d, iters));
     best[2] = fminf(best[2], run<2>(zi, zt, out, b, n, d, iters));
     best[3] = fminf(best[3], run<3>(zi, zt, out, b, n, d, iters));
     best[4] = fminf(best[4], run<4>(zi, zt, out, b, n, d, iters));
+    best[5] = fminf(best[5], run<5>(zi, zt, out, b, n, d, iters));
+    best[6] = fminf(best[6], run<6>(zi, zt, out, b, n, d, iters));
   }
-  for (int v = 0; v < 5; ++v)
+  for (int v = 0; v < 7; ++v)
     printf("%-16s %8.3f ms  %7.1f TF/s\n", names[v], best[v],
            flops / best[v] / 1e9);
   return 0;
