@@ -113,7 +113,7 @@ __launch_bounds__(THREADS) __global__ void fwd_kernel(
     for (int j = 0; j < 4; ++j) {
       uint4 va_ = *reinterpret_cast<const uint4*>(abase + va[j]);
       uint4 vb_ = *reinterpret_cast<const uint4*>(bbase + va[j]);
-      asm volatile("" :: "v"(va_), "v"(vb_));
+      asm volatile("" :: "v"(va_.x), "v"(va_.w), "v"(vb_.x), "v"(vb_.w));
     }
     abase += 128;
     bbase += 128;
