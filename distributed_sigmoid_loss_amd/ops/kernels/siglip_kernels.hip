@@ -54,6 +54,7 @@
 // Compile: hipcc --offload-arch=gfx950 -O3 -shared -fPIC.
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_fp8.h>
 #include <cstdint>
 #include <climits>
 
@@ -179,7 +180,9 @@ __device__ __forceinline__ void mma_ktile_fp8(const char* smem,
   }
 }
 
-template <int MODE, bool INTERIOR, int EB>
+// g-slab element bytes follow the compute dtype: bf16 kernels emit bf16 g,
+// fp8 kernels emit e4m3 g (×448 fixed scale).
+template <int MODE, bool INTERIOR, int EB, int EB_G = EB>
 __device__ __forceinline__ void tile_body(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     float t, float bias, float* __restrict__ out, __bf16* __restrict__ g_out,
@@ -343,8 +346,12 @@ __device__ __forceinline__ void tile_body(
     // One per-lane base offset + per-(mi,reg) scalar row offset keeps the
     // store addressing affine — per-element (size_t)grow*n math made the
     // allocator hoist 128 addresses and spill.  Caller guarantees
-    // b*n*2 < 2^32 (ops/__init__.py column-chunks the slab).
-    __bf16* gb = g_out + (size_t)row_base * n + col_base;
+    // b*n*esz < 2^32 (ops/__init__.py column-chunks the slab).
+    // bf16 path writes a bf16 slab; fp8 path writes e4m3 at a FIXED ×448
+    // scale (|g| ≤ 1 by construction, so max maps to e4m3's max normal) —
+    // consumed by torch._scaled_mm with scale 1/448 folded in.
+    char* gb = reinterpret_cast<char*>(g_out) +
+        ((size_t)row_base * n + col_base) * EB_G;
     const unsigned lane_off =
         (unsigned)(wrow + (lane >> 4) * 4) * (unsigned)n
         + (unsigned)(wcol + (lane & 15));
@@ -363,7 +370,12 @@ __device__ __forceinline__ void tile_body(
             const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
             const float gv = sigmoid_fast(pos ? z : -z);
             const float g = pos ? -gv : gv;
-            gb[lane_off + row_off + ni * 16] = (__bf16)g;
+            const unsigned off = lane_off + row_off + ni * 16;
+            if (EB_G == 1)
+              reinterpret_cast<unsigned char*>(gb)[off] =
+                  __hip_fp8_e4m3(g * 448.0f).__x;
+            else
+              reinterpret_cast<__bf16*>(gb)[off] = (__bf16)g;
             s0 += g * dot;
             s1 += g;
           }
