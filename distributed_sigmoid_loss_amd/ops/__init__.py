@@ -66,7 +66,7 @@ def _load():
     lib.siglip_fwd_fp8.restype = ctypes.c_int
     lib.siglip_fwd_fp8.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] * 5
     lib.siglip_bwd_g_fp8.restype = ctypes.c_int
-    lib.siglip_bwd_g_fp8.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] * 5
+    lib.siglip_bwd_g_fp8.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5
     _lib = lib
     return _lib
 
@@ -205,6 +205,9 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     # must be 16-aligned; otherwise dequantize and use rocBLAS.
     use_mm8 = (quant == "fp8" and b % 16 == 0 and n % 16 == 0
                and d % 16 == 0 and hasattr(torch, "_scaled_mm"))
+    fp8_ok = quant == "fp8" and b % 4 == 0   # packed gt stores need b%4
+    if quant == "fp8" and not fp8_ok:
+        raise RuntimeError("fp8 backward requires batch % 4 == 0")
 
     def mm8(a8, b8_rowmajor, s_ab, out_rows):
         """(out_rows, d) = a8 @ b8 with b8 given row-major (k, d)."""
@@ -213,28 +216,45 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                                 scale_b=torch.ones((), device=dev),
                                 out_dtype=torch.bfloat16)
 
-    def run_g(j0, j1, g_slab, diag):
+    def run_g(j0, j1, g_slab, diag, gt_slab=None):
         zt_slab = zt_k[j0:j1]
-        _check(g_fn(
-            ctypes.c_void_p(stream),
-            ctypes.c_void_p(zi_k.data_ptr()),
-            ctypes.c_void_p(zt_slab.data_ptr()),
-            ctypes.c_void_p(tp_k.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
-            ctypes.c_void_p(g_slab.data_ptr()),
-            ctypes.c_void_p(scal.data_ptr()),
-            b, j1 - j0, d, diag, _kernel_flags()),
-            "siglip_bwd_g")
+        if quant == "fp8":
+            _check(g_fn(
+                ctypes.c_void_p(stream),
+                ctypes.c_void_p(zi_k.data_ptr()),
+                ctypes.c_void_p(zt_slab.data_ptr()),
+                ctypes.c_void_p(tp_k.data_ptr()),
+                ctypes.c_void_p(bp.data_ptr()),
+                ctypes.c_void_p(g_slab.data_ptr()),
+                ctypes.c_void_p(gt_slab.data_ptr()),
+                ctypes.c_void_p(scal.data_ptr()),
+                b, j1 - j0, d, diag, _kernel_flags()),
+                "siglip_bwd_g_fp8")
+        else:
+            _check(g_fn(
+                ctypes.c_void_p(stream),
+                ctypes.c_void_p(zi_k.data_ptr()),
+                ctypes.c_void_p(zt_slab.data_ptr()),
+                ctypes.c_void_p(tp_k.data_ptr()),
+                ctypes.c_void_p(bp.data_ptr()),
+                ctypes.c_void_p(g_slab.data_ptr()),
+                ctypes.c_void_p(scal.data_ptr()),
+                b, j1 - j0, d, diag, _kernel_flags()),
+                "siglip_bwd_g_bf16")
 
     if step >= n:
         # Single-slab fast path: no fp32 accumulation round trips.  dztxt is
         # produced FIRST so the caller's on_dztxt hook (e.g. an async RCCL
         # reduce-scatter) overlaps with the dzimg GEMM below.
         g = torch.empty((b, n), device=dev, dtype=g_dtype)
-        run_g(0, n, g, _DIAG_NONE if diag_offset is None else int(diag_offset))
+        gt = (torch.empty((n, b), device=dev, dtype=g_dtype)
+              if quant == "fp8" else None)
+        run_g(0, n, g, _DIAG_NONE if diag_offset is None else int(diag_offset),
+              gt)
         if use_mm8:
             s_t = ((scale / 448.0) * st).reshape(())
             s_i = ((scale / 448.0) * si).reshape(())
-            dztxt = mm8(g.t().contiguous(), zi_k, s_i, n)
+            dztxt = mm8(gt, zi_k, s_i, n)
             if on_dztxt is not None:
                 on_dztxt(dztxt)
             dzimg = mm8(g, zt_k, s_t, b)
@@ -249,18 +269,24 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         dzimg_acc = torch.zeros((b, d), device=dev, dtype=torch.float32)
         dztxt = torch.empty((n, d), device=dev, dtype=torch.bfloat16)
         g_buf = torch.empty((b, step), device=dev, dtype=g_dtype)
+        gt_buf = (torch.empty((step, b), device=dev, dtype=g_dtype)
+                  if quant == "fp8" else None)
         for j0 in range(0, n, step):
             j1 = min(j0 + step, n)
             c = j1 - j0
             g = g_buf if c == step else torch.empty(
                 (b, c), device=dev, dtype=g_dtype)
+            gt = None
+            if quant == "fp8":
+                gt = gt_buf if c == step else torch.empty(
+                    (c, b), device=dev, dtype=g_dtype)
             diag = _DIAG_NONE if diag_offset is None else int(diag_offset) - j0
-            run_g(j0, j1, g, diag)
+            run_g(j0, j1, g, diag, gt)
             if use_mm8 and c % 16 == 0:
                 s_t = ((scale / 448.0) * st).reshape(())
                 s_i = ((scale / 448.0) * si).reshape(())
                 dzimg_acc += mm8(g, zt_k[j0:j1], s_t, b).float()
-                dztxt[j0:j1] = mm8(g.t().contiguous(), zi_k, s_i, c)
+                dztxt[j0:j1] = mm8(gt, zi_k, s_i, c)
             else:
                 g16 = (g.to(torch.bfloat16) * (1.0 / 448.0)
                        ) if quant == "fp8" else g

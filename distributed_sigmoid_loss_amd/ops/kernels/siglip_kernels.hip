@@ -186,6 +186,7 @@ template <int MODE, bool INTERIOR, int EB, int EB_G = EB>
 __device__ __forceinline__ void tile_body(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     float t, float bias, float* __restrict__ out, __bf16* __restrict__ g_out,
+    unsigned char* __restrict__ gt_out,
     int b, int n, int d, int diag, int row_base, int col_base, char* smem) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -347,22 +348,30 @@ __device__ __forceinline__ void tile_body(
     // store addressing affine — per-element (size_t)grow*n math made the
     // allocator hoist 128 addresses and spill.  Caller guarantees
     // b*n*esz < 2^32 (ops/__init__.py column-chunks the slab).
-    // bf16 path writes a bf16 slab; fp8 path writes e4m3 at a FIXED ×448
-    // scale (|g| ≤ 1 by construction, so max maps to e4m3's max normal) —
-    // consumed by torch._scaled_mm with scale 1/448 folded in.
+    // bf16 path writes one bf16 slab; fp8 path writes e4m3 at a FIXED ×448
+    // scale (|g| ≤ 1 by construction) AND a transposed (n, b) slab — both
+    // consumed by torch._scaled_mm (mat1 must be row-major; a Python-side
+    // 1-GB fp8 transpose measured ~5 ms, the fused stores are free).  The
+    // 4 accumulator regs of a fragment are 4 consecutive gᵀ columns → one
+    // packed 4-byte store.
     char* gb = reinterpret_cast<char*>(g_out) +
         ((size_t)row_base * n + col_base) * EB_G;
     const unsigned lane_off =
         (unsigned)(wrow + (lane >> 4) * 4) * (unsigned)n
         + (unsigned)(wcol + (lane & 15));
+    unsigned char* gtb = gt_out +
+        ((size_t)col_base * b + row_base);
+    const unsigned lane_off_t =
+        (unsigned)(wcol + (lane & 15)) * (unsigned)b
+        + (unsigned)(wrow + (lane >> 4) * 4);
 #pragma unroll
     for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)n;
-        const int grow = row_base + wrow + mi * 16 + (lane >> 4) * 4 + reg;
+      for (int ni = 0; ni < FN; ++ni) {
+        unsigned packed = 0;
 #pragma unroll
-        for (int ni = 0; ni < FN; ++ni) {
+        for (int reg = 0; reg < 4; ++reg) {
+          const int grow = row_base + wrow + mi * 16 + (lane >> 4) * 4 + reg;
           const int gcol = col_base + wcol + ni * 16 + (lane & 15);
           if (INTERIOR || (grow < b && gcol < n)) {
             const float dot = acc[mi][ni][reg];
@@ -370,14 +379,33 @@ __device__ __forceinline__ void tile_body(
             const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
             const float gv = sigmoid_fast(pos ? z : -z);
             const float g = pos ? -gv : gv;
-            const unsigned off = lane_off + row_off + ni * 16;
-            if (EB_G == 1)
-              reinterpret_cast<unsigned char*>(gb)[off] =
-                  __hip_fp8_e4m3(g * 448.0f).__x;
-            else
-              reinterpret_cast<__bf16*>(gb)[off] = (__bf16)g;
+            const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)n;
+            if (EB_G == 1) {
+              const unsigned char q = __hip_fp8_e4m3(g * 448.0f).__x;
+              reinterpret_cast<unsigned char*>(gb)[
+                  lane_off + row_off + ni * 16] = q;
+              packed |= (unsigned)q << (8 * reg);
+            } else {
+              reinterpret_cast<__bf16*>(gb)[lane_off + row_off + ni * 16] =
+                  (__bf16)g;
+            }
             s0 += g * dot;
             s1 += g;
+          }
+        }
+        if (EB_G == 1) {
+          const unsigned toff = lane_off_t + (unsigned)(ni * 16) * (unsigned)b
+              + (unsigned)(mi * 16);
+          if (INTERIOR) {
+            *reinterpret_cast<unsigned*>(gtb + toff) = packed;
+          } else {
+            const int grow0 = row_base + wrow + mi * 16 + (lane >> 4) * 4;
+            const int gcol = col_base + wcol + ni * 16 + (lane & 15);
+            if (gcol < n)
+#pragma unroll
+              for (int reg = 0; reg < 4; ++reg)
+                if (grow0 + reg < b)
+                  gtb[toff + reg] = (unsigned char)(packed >> (8 * reg));
           }
         }
       }
@@ -425,14 +453,15 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
     float* __restrict__ out, __bf16* __restrict__ g_out,
+    unsigned char* __restrict__ gt_out,
     int b, int n, int d, int diag, int flags) {
   __shared__ char smem[4 * TILE_BYTES];
   int bx, by;
   remap_block(flags, bx, by);
   const float t = __expf(*t_prime);
   const float bias = *bias_p;
-  tile_body<MODE, true, EB>(zimg, ztxt, t, bias, out, g_out, b, n, d, diag,
-                            bx * BM, by * BN, smem);
+  tile_body<MODE, true, EB>(zimg, ztxt, t, bias, out, g_out, gt_out,
+                            b, n, d, diag, bx * BM, by * BN, smem);
 }
 
 // General kernel: interior blocks take the DMA fast path, edge blocks the
@@ -443,6 +472,7 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
     float* __restrict__ out, __bf16* __restrict__ g_out,
+    unsigned char* __restrict__ gt_out,
     int b, int n, int d, int diag, int flags) {
   __shared__ char smem[4 * TILE_BYTES];
   int bx, by;
@@ -456,11 +486,11 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
   const bool interior = (row_base + BM <= b) && (col_base + BN <= n) &&
       (d % (128 / EB) == 0) && (n % 8 == 0);
   if (interior)
-    tile_body<MODE, true, EB>(zimg, ztxt, t, bias, out, g_out, b, n, d, diag,
-                              row_base, col_base, smem);
+    tile_body<MODE, true, EB>(zimg, ztxt, t, bias, out, g_out, gt_out,
+                              b, n, d, diag, row_base, col_base, smem);
   else
-    tile_body<MODE, false, EB>(zimg, ztxt, t, bias, out, g_out, b, n, d,
-                               diag, row_base, col_base, smem);
+    tile_body<MODE, false, EB>(zimg, ztxt, t, bias, out, g_out, gt_out,
+                               b, n, d, diag, row_base, col_base, smem);
 }
 
 inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
@@ -468,7 +498,7 @@ inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 template <int MODE, int EB>
 int launch(uintptr_t stream, const void* zimg, const void* ztxt,
            const void* t_prime, const void* bias, void* out, void* g_out,
-           int b, int n, int d, int diag, int flags) {
+           void* gt_out, int b, int n, int d, int diag, int flags) {
   if (b <= 0 || n <= 0) return (int)hipSuccess;
   if (d % (16 / EB) != 0) return (int)hipErrorInvalidValue;
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
@@ -479,13 +509,15 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
                        dim3(THREADS), 0, (hipStream_t)stream,
                        (const char*)zimg, (const char*)ztxt,
                        (const float*)t_prime, (const float*)bias,
-                       (float*)out, (__bf16*)g_out, b, n, d, diag, flags);
+                       (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
+                       b, n, d, diag, flags);
   else
     hipLaunchKernelGGL((siglip_tile_kernel<MODE, EB>), grid, dim3(THREADS),
                        0, (hipStream_t)stream,
                        (const char*)zimg, (const char*)ztxt,
                        (const float*)t_prime, (const float*)bias,
-                       (float*)out, (__bf16*)g_out, b, n, d, diag, flags);
+                       (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
+                       b, n, d, diag, flags);
   return (int)hipGetLastError();
 }
 
@@ -499,28 +531,31 @@ int siglip_fwd_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                     const void* t_prime, const void* bias, void* loss_out,
                     int b, int n, int d, int diag, int flags) {
   return launch<0, 2>(stream, zimg, ztxt, t_prime, bias, loss_out, nullptr,
-                      b, n, d, diag, flags);
+                      nullptr, b, n, d, diag, flags);
 }
 
 int siglip_bwd_g_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                       const void* t_prime, const void* bias, void* g_out,
                       void* scal, int b, int n, int d, int diag, int flags) {
   return launch<1, 2>(stream, zimg, ztxt, t_prime, bias, scal, g_out,
-                      b, n, d, diag, flags);
+                      nullptr, b, n, d, diag, flags);
 }
 
 int siglip_fwd_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
                    const void* t_prime, const void* bias, void* loss_out,
                    int b, int n, int d, int diag, int flags) {
   return launch<0, 1>(stream, zimg, ztxt, t_prime, bias, loss_out, nullptr,
-                      b, n, d, diag, flags);
+                      nullptr, b, n, d, diag, flags);
 }
 
+// fp8 backward emits g (b,n) AND its transpose gt (n,b), both e4m3 ×448.
 int siglip_bwd_g_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
                      const void* t_prime, const void* bias, void* g_out,
-                     void* scal, int b, int n, int d, int diag, int flags) {
+                     void* gt_out, void* scal, int b, int n, int d, int diag,
+                     int flags) {
+  if (b % 4 != 0) return (int)hipErrorInvalidValue;  // packed 4-B gt stores
   return launch<1, 1>(stream, zimg, ztxt, t_prime, bias, scal, g_out,
-                      b, n, d, diag, flags);
+                      gt_out, b, n, d, diag, flags);
 }
 
 }  // extern "C"
