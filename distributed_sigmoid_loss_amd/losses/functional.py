@@ -156,9 +156,12 @@ class _FusedSigmoidLoss(torch.autograd.Function):
         from .. import ops
         zimg = zimg.contiguous()
         ztxt = ztxt.contiguous()
+        qc = (ops.quantize_fp8_pair(zimg, ztxt)
+              if (quant == "fp8" and zimg.is_cuda) else None)
         loss = ops.siglip_fwd(zimg, ztxt, t_prime, bias, diag_offset,
-                              quant=quant)
-        ctx.save_for_backward(zimg, ztxt, t_prime, bias)
+                              quant=quant, qcache=qc)
+        saved = (zimg, ztxt, t_prime, bias) + (qc if qc is not None else ())
+        ctx.save_for_backward(*saved)
         ctx.diag_offset = diag_offset
         ctx.col_chunk = col_chunk   # None → single slab when addressable
         ctx.quant = quant
@@ -167,10 +170,11 @@ class _FusedSigmoidLoss(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_output):
         from .. import ops
-        zimg, ztxt, t_prime, bias = ctx.saved_tensors
+        zimg, ztxt, t_prime, bias = ctx.saved_tensors[:4]
+        qc = ctx.saved_tensors[4:] if len(ctx.saved_tensors) > 4 else None
         dzimg, dztxt, dt_prime, dbias = ops.siglip_bwd(
             zimg, ztxt, t_prime, bias, ctx.diag_offset, grad_output,
-            ctx.col_chunk, quant=ctx.quant)
+            ctx.col_chunk, quant=ctx.quant, qcache=qc)
         return dzimg, dztxt, dt_prime, dbias, None, None, None
 
 

@@ -121,9 +121,18 @@ def _quant_fp8(x: torch.Tensor):
     return q, scale
 
 
+def quantize_fp8_pair(zimg: torch.Tensor, ztxt: torch.Tensor):
+    """Quantize both embedding tensors once; pass the result as ``qcache`` to
+    both :func:`siglip_fwd` and :func:`siglip_bwd` so an fwd+bwd step pays a
+    single quantization pass."""
+    zi_q, si = _quant_fp8(zimg)
+    zt_q, st = _quant_fp8(ztxt)
+    return zi_q, si, zt_q, st
+
+
 def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                bias: torch.Tensor, diag_offset: Optional[int],
-               quant: str = "bf16") -> torch.Tensor:
+               quant: str = "bf16", qcache=None) -> torch.Tensor:
     lib = _require_lib()
     _validate(zimg, ztxt, quant)
     b, d = zimg.shape
@@ -132,8 +141,8 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     tp = _prep_scalar(t_prime, dev)
     bp = _prep_scalar(bias, dev)
     if quant == "fp8":
-        zi_q, si = _quant_fp8(zimg)
-        zt_q, st = _quant_fp8(ztxt)
+        zi_q, si, zt_q, st = (qcache if qcache is not None
+                              else quantize_fp8_pair(zimg, ztxt))
         tp = tp + si.log() + st.log()
         zi_ptr, zt_ptr = zi_q.data_ptr(), zt_q.data_ptr()
         fn = lib.siglip_fwd_fp8
@@ -155,7 +164,7 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
 def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                bias: torch.Tensor, diag_offset: Optional[int],
                grad_output: torch.Tensor, col_chunk: Optional[int],
-               quant: str = "bf16", on_dztxt=None):
+               quant: str = "bf16", on_dztxt=None, qcache=None):
     """Returns (dzimg, dztxt, dt_prime, dbias).
 
     Per column slab: the fused kernel recomputes logit tiles (MFMA) and writes
@@ -171,10 +180,11 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     bp = _prep_scalar(bias, dev)
     t_true = tp.exp()
     if quant == "fp8":
-        # Same per-tensor scales as forward (identical inputs → identical
-        # amax); the g kernel sees t_eff so its logits match the forward's.
-        zi_q, si = _quant_fp8(zimg)
-        zt_q, st = _quant_fp8(ztxt)
+        # Reuse the forward's quantization when provided (identical inputs
+        # give identical amax, so recomputing is equivalent but wasteful);
+        # the g kernel sees t_eff so its logits match the forward's.
+        zi_q, si, zt_q, st = (qcache if qcache is not None
+                              else quantize_fp8_pair(zimg, ztxt))
         tp_k = tp + si.log() + st.log()
         g_fn = lib.siglip_bwd_g_fp8
         zi_k, zt_k = zi_q, zt_q
