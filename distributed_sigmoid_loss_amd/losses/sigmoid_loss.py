@@ -251,6 +251,28 @@ class SigLipLoss(nn.Module):
         self.impl = impl
         self.col_chunk = col_chunk
 
+    def get_ground_truth(self, device, dtype, num_logits,
+                         negative_only=False) -> torch.Tensor:
+        """Materialized ±1 label block — API parity with the reference
+        (``rwightman_sigmoid_loss.py:43-47``).  The fused path never calls
+        this (labels are an index predicate); it exists for users porting
+        code that consumes it."""
+        labels = -torch.ones((num_logits, num_logits), device=device,
+                             dtype=dtype)
+        if not negative_only:
+            labels = 2 * torch.eye(num_logits, device=device,
+                                   dtype=dtype) + labels
+        return labels
+
+    def get_logits(self, image_features, text_features, logit_scale,
+                   logit_bias=None):
+        """Pairwise logits — API parity with the reference
+        (``rwightman_sigmoid_loss.py:49-53``)."""
+        logits = logit_scale.exp() * image_features @ text_features.T
+        if logit_bias is not None:
+            logits = logits + logit_bias
+        return logits
+
     def _loss(self, image_features, text_features, logit_scale, logit_bias,
               negative_only=False):
         off = None if negative_only else 0

@@ -80,3 +80,24 @@ def test_output_dict_form():
 def test_horovod_unsupported():
     with pytest.raises(NotImplementedError):
         SigLipLoss(use_horovod=True)
+
+
+def test_get_ground_truth_and_logits_api():
+    """API-parity methods (reference rwightman_sigmoid_loss.py:43-53)."""
+    mod = SigLipLoss(rank=0, world_size=1)
+    lab = mod.get_ground_truth("cpu", torch.float32, 4)
+    assert lab.shape == (4, 4)
+    assert torch.all(lab.diagonal() == 1) and lab.sum() == 4 - 12
+    lab_neg = mod.get_ground_truth("cpu", torch.float32, 4,
+                                   negative_only=True)
+    assert torch.all(lab_neg == -1)
+
+    torch.manual_seed(0)
+    zi = torch.randn(3, 8)
+    zt = torch.randn(5, 8)
+    scale = torch.tensor(0.5)
+    bias = torch.tensor(-2.0)
+    logits = mod.get_logits(zi, zt, scale, bias)
+    assert torch.allclose(logits, scale.exp() * zi @ zt.T + bias)
+    assert torch.allclose(mod.get_logits(zi, zt, scale),
+                          scale.exp() * zi @ zt.T)
