@@ -38,10 +38,11 @@ from typing import Optional
 import torch
 import torch.nn.functional as F
 
-# Default column-slab size for the backward g-buffer: bounds workspace to
-# b × DEFAULT_COL_CHUNK bf16 elements (e.g. 131072 × 8192 × 2 B = 2 GiB,
-# comfortably inside 288 GB HBM3E alongside the embeddings).
-DEFAULT_COL_CHUNK = 8192
+# With col_chunk=None the GPU backward uses a single g slab whenever its
+# 32-bit store addressing allows (b·n·esz < 2^32) and auto-halves the slab
+# otherwise — e.g. per-GPU b=131072 against n=131072 runs at 8192-column
+# slabs (2 GiB workspace), comfortably inside 288 GB HBM3E.  Set col_chunk
+# explicitly to bound workspace tighter.
 
 
 def _labels(b: int, n: int, diag_offset: Optional[int], device, dtype):
@@ -194,7 +195,8 @@ def sigmoid_contrastive_loss(zimg: torch.Tensor, ztxt: torch.Tensor,
         bias: scalar additive logit bias.
         diag_offset: column of row 0's positive pair, or ``None`` for a
             negatives-only block (remote chunk).
-        col_chunk: column-slab size bounding workspace (default 8192 on GPU).
+        col_chunk: column-slab size bounding workspace; None = single slab
+            when addressable (see module note above).
         impl: ``auto`` | ``hip`` | ``torch``.
 
     Returns a scalar tensor (caller applies the ``1/b`` normalization).
