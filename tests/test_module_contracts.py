@@ -14,8 +14,8 @@ from distributed_sigmoid_loss_amd import DistributedSigmoidLoss, SigLipLoss
 
 def test_state_dict_roundtrip():
     mod = DistributedSigmoidLoss(8)
-    assert math.isclose(float(mod.t_prime), math.log(10.0), rel_tol=1e-6)
-    assert float(mod.bias) == -10.0
+    assert math.isclose(float(mod.t_prime.detach()), math.log(10.0), rel_tol=1e-6)
+    assert float(mod.bias.detach()) == -10.0
     with torch.no_grad():
         mod.t_prime.fill_(1.5)
         mod.bias.fill_(-3.25)
@@ -24,8 +24,8 @@ def test_state_dict_roundtrip():
 
     mod2 = DistributedSigmoidLoss(8)
     mod2.load_state_dict(sd)
-    assert float(mod2.t_prime) == 1.5
-    assert float(mod2.bias) == -3.25
+    assert float(mod2.t_prime.detach()) == 1.5
+    assert float(mod2.bias.detach()) == -3.25
 
 
 def test_loss_params_in_optimizer():
@@ -33,10 +33,10 @@ def test_loss_params_in_optimizer():
     opt = torch.optim.SGD(mod.parameters(), lr=0.1)
     zi = F.normalize(torch.randn(4, 16), dim=-1)
     zt = F.normalize(torch.randn(4, 16), dim=-1)
-    before = (float(mod.t_prime), float(mod.bias))
+    before = (float(mod.t_prime.detach()), float(mod.bias.detach()))
     mod(zi, zt).backward()
     opt.step()
-    after = (float(mod.t_prime), float(mod.bias))
+    after = (float(mod.t_prime.detach()), float(mod.bias.detach()))
     assert before != after  # both params updated through the optimizer
 
 
