@@ -93,9 +93,9 @@ def main():
         if rank == 0 and (step + 1) % args.log_every == 0:
             phases = "  ".join(f"{k}={v:.2f}ms"
                                for k, v in timer.summary().items())
-            print(f"step {step + 1:4d}  loss={float(loss):.4f}  "
-                  f"t={float(loss_mod.t_prime.exp()):.3f}  "
-                  f"bias={float(loss_mod.bias):.3f}  {phases}", flush=True)
+            print(f"step {step + 1:4d}  loss={float(loss.detach()):.4f}  "
+                  f"t={float(loss_mod.t_prime.detach().exp()):.3f}  "
+                  f"bias={float(loss_mod.bias.detach()):.3f}  {phases}", flush=True)
             timer.rows.clear()
 
     if device == "cuda":
