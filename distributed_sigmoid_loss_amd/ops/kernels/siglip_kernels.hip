@@ -382,12 +382,17 @@ __device__ __forceinline__ void tile_body(
             const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)n;
             if (EB_G == 1) {
               const unsigned char q = __hip_fp8_e4m3(g * 448.0f).__x;
-              reinterpret_cast<unsigned char*>(gb)[
-                  lane_off + row_off + ni * 16] = q;
+              __builtin_nontemporal_store(
+                  q, reinterpret_cast<unsigned char*>(gb) +
+                      lane_off + row_off + ni * 16);
               packed |= (unsigned)q << (8 * reg);
             } else {
-              reinterpret_cast<__bf16*>(gb)[lane_off + row_off + ni * 16] =
-                  (__bf16)g;
+              // Non-temporal: the slab is written once and read back by the
+              // GEMMs from HBM anyway (2 GB ≫ L2); keeping it out of the
+              // caches protects the staging DMA's hit rate.
+              __builtin_nontemporal_store(
+                  (__bf16)g, reinterpret_cast<__bf16*>(gb) +
+                      lane_off + row_off + ni * 16);
             }
             s0 += g * dot;
             s1 += g;
@@ -397,7 +402,8 @@ __device__ __forceinline__ void tile_body(
           const unsigned toff = lane_off_t + (unsigned)(ni * 16) * (unsigned)b
               + (unsigned)(mi * 16);
           if (INTERIOR) {
-            *reinterpret_cast<unsigned*>(gtb + toff) = packed;
+            __builtin_nontemporal_store(
+                packed, reinterpret_cast<unsigned*>(gtb + toff));
           } else {
             const int grow0 = row_base + wrow + mi * 16 + (lane >> 4) * 4;
             const int gcol = col_base + wcol + ni * 16 + (lane & 15);
