@@ -382,17 +382,12 @@ __device__ __forceinline__ void tile_body(
             const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)n;
             if (EB_G == 1) {
               const unsigned char q = __hip_fp8_e4m3(g * 448.0f).__x;
-              __builtin_nontemporal_store(
-                  q, reinterpret_cast<unsigned char*>(gb) +
-                      lane_off + row_off + ni * 16);
+              reinterpret_cast<unsigned char*>(gb)[
+                  lane_off + row_off + ni * 16] = q;
               packed |= (unsigned)q << (8 * reg);
             } else {
-              // Non-temporal: the slab is written once and read back by the
-              // GEMMs from HBM anyway (2 GB ≫ L2); keeping it out of the
-              // caches protects the staging DMA's hit rate.
-              __builtin_nontemporal_store(
-                  (__bf16)g, reinterpret_cast<__bf16*>(gb) +
-                      lane_off + row_off + ni * 16);
+              reinterpret_cast<__bf16*>(gb)[lane_off + row_off + ni * 16] =
+                  (__bf16)g;
             }
             s0 += g * dot;
             s1 += g;
@@ -402,8 +397,7 @@ __device__ __forceinline__ void tile_body(
           const unsigned toff = lane_off_t + (unsigned)(ni * 16) * (unsigned)b
               + (unsigned)(mi * 16);
           if (INTERIOR) {
-            __builtin_nontemporal_store(
-                packed, reinterpret_cast<unsigned*>(gtb + toff));
+            *reinterpret_cast<unsigned*>(gtb + toff) = packed;
           } else {
             const int grow0 = row_base + wrow + mi * 16 + (lane >> 4) * 4;
             const int gcol = col_base + wcol + ni * 16 + (lane & 15);
