@@ -265,7 +265,9 @@ __device__ __forceinline__ void tile_body(
         } else {
           asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
         }
+        __builtin_amdgcn_s_setprio(1);
         mma_half_bf16(smem, aAddr[0], bAddr[0], acc);
+        __builtin_amdgcn_s_setprio(0);
         if (more) {
           stage_half((kt + 1) & 1, 1);
           advance();
@@ -273,7 +275,9 @@ __device__ __forceinline__ void tile_body(
         } else {
           asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
         }
+        __builtin_amdgcn_s_setprio(1);
         mma_half_bf16(smem, aAddr[1], bAddr[1], acc);
+        __builtin_amdgcn_s_setprio(0);
         aAddr[0] ^= 2 * TILE_BYTES; aAddr[1] ^= 2 * TILE_BYTES;
         bAddr[0] ^= 2 * TILE_BYTES; bAddr[1] ^= 2 * TILE_BYTES;
         // No end-of-iteration barrier: each half-region's write is already
