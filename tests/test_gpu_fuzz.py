@@ -1,0 +1,88 @@
+"""Seeded shape-fuzz of the fused kernels vs the fp32 PyTorch reference.
+
+Catches tile-boundary and addressing edge cases the fixed-shape tests miss
+(ragged b/n, short/tall blocks, odd diag offsets, K tails).  Seeded, small,
+fast (~10 s) so it stays in the round-end GPU suite.
+"""
+
+import math
+import random
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():  # pragma: no cover
+    pytest.skip("needs a GPU", allow_module_level=True)
+
+from distributed_sigmoid_loss_amd import ops
+from distributed_sigmoid_loss_amd.losses.functional import (
+    _torch_loss,
+    _torch_bwd,
+)
+
+
+def _cases(n_cases, seed, d_mult):
+    rng = random.Random(seed)
+    cases = []
+    for _ in range(n_cases):
+        b = rng.randrange(1, 600)
+        n = rng.randrange(1, 600)
+        d = d_mult * rng.randrange(1, 96 // d_mult + 1) * 8 // 8
+        d = max(d_mult, (d // d_mult) * d_mult)
+        diag = rng.choice([None, 0, rng.randrange(-b, n + 1)])
+        cases.append((b, n, d, diag))
+    return cases
+
+
+@pytest.mark.parametrize("case", _cases(8, seed=1234, d_mult=8))
+def test_fuzz_bf16_fwd_bwd(case):
+    b, n, d, diag = case
+    g = torch.Generator().manual_seed(hash(case) & 0xFFFF)
+    zi = F.normalize(torch.randn(b, d, generator=g), dim=-1) \
+        .cuda().bfloat16()
+    zt = F.normalize(torch.randn(n, d, generator=g), dim=-1) \
+        .cuda().bfloat16()
+    tp = torch.tensor(math.log(7.0), device="cuda")
+    bs = torch.tensor(-6.0, device="cuda")
+    go = torch.tensor(0.9, device="cuda")
+
+    got = ops.siglip_fwd(zi, zt, tp, bs, diag)
+    want = _torch_loss(zi.float(), zt.float(), tp.float(), bs.float(), diag,
+                       None)
+    torch.cuda.synchronize()
+    assert torch.allclose(got, want, rtol=2e-2, atol=2e-2), \
+        (case, got.item(), want.item())
+
+    dzi, dzt, dtp, dbs = ops.siglip_bwd(zi, zt, tp, bs, diag, go, None)
+    r = _torch_bwd(zi.float(), zt.float(), tp.float(), bs.float(), diag,
+                   go.float(), None)
+    torch.cuda.synchronize()
+    assert torch.allclose(dzi.float(), r[0], rtol=6e-2, atol=2e-3), case
+    assert torch.allclose(dzt.float(), r[1], rtol=6e-2, atol=2e-3), case
+    assert torch.allclose(dtp.float(), r[2], rtol=3e-2, atol=2e-3), case
+    assert torch.allclose(dbs.float(), r[3], rtol=3e-2, atol=2e-3), case
+
+
+@pytest.mark.parametrize("case", _cases(4, seed=77, d_mult=16))
+def test_fuzz_fp8_fwd(case):
+    b, n, d, diag = case
+    b = max(4, (b // 4) * 4)   # fp8 backward requires b % 4; fwd any b
+    g = torch.Generator().manual_seed(hash(case) & 0xFFFF)
+    zi = F.normalize(torch.randn(b, d, generator=g), dim=-1) \
+        .cuda().bfloat16()
+    zt = F.normalize(torch.randn(n, d, generator=g), dim=-1) \
+        .cuda().bfloat16()
+    tp = torch.tensor(math.log(7.0), device="cuda")
+    bs = torch.tensor(-6.0, device="cuda")
+
+    got = ops.siglip_fwd(zi, zt, tp, bs, diag, quant="fp8")
+    qz = ops._quant_fp8(zi)
+    qt = ops._quant_fp8(zt)
+    want = _torch_loss(qz[0].float() * qz[1], qt[0].float() * qt[1],
+                       tp.float(), bs.float(), diag, None)
+    torch.cuda.synchronize()
+    assert torch.allclose(got, want, rtol=3e-2, atol=3e-2), \
+        (case, got.item(), want.item())
