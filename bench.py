@@ -22,6 +22,7 @@ from __future__ import annotations
 
 import argparse
 import json
+import sys
 import time
 
 import torch
@@ -54,6 +55,10 @@ def parse_args():
                    help="column slab size for chunked negatives")
     p.add_argument("--csv", default=None,
                    help="write per-step phase timings (ms) to this CSV")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the steady-state step in a hipGraph and "
+                        "replay it (single-GPU; falls back to eager if "
+                        "capture fails)")
     p.add_argument("--ddp", action="store_true",
                    help="wrap the towers in torch DDP (bucketed RCCL "
                         "all-reduce overlapped with backward) instead of "
@@ -92,8 +97,12 @@ def main():
     img_feats = torch.randn(b, args.dim, device=device, dtype=dtype)
     txt_feats = torch.randn(b, args.dim, device=device, dtype=dtype)
 
+    use_graph = (args.graph and device == "cuda" and world == 1
+                 and args.csv is None)
+
     def step():
-        opt.zero_grad(set_to_none=True)
+        # Graph mode needs stable grad buffers across replays.
+        opt.zero_grad(set_to_none=not use_graph)
         with timer.phase("encode"):
             zi, zt = model(img_feats, txt_feats)
         with timer.phase("loss_fwd"):
@@ -118,10 +127,23 @@ def main():
 
     for _ in range(args.warmup):
         step()
+    run_step = step
+    if use_graph:
+        try:
+            graph = torch.cuda.CUDAGraph()
+            torch.cuda.synchronize()
+            with torch.cuda.graph(graph):
+                step()
+            run_step = graph.replay
+        except Exception as e:  # pragma: no cover - capture support varies
+            print(f"# graph capture failed, falling back to eager: {e}",
+                  file=sys.stderr)
+            run_step = step
+        run_step()   # one replay as extra warmup
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        run_step()
     sync()
     elapsed = time.perf_counter() - t0
 
@@ -160,6 +182,7 @@ def main():
                 "seq_len": None,
                 "parallelism": f"dp{world}-{args.strategy}",
                 "impl": args.impl,
+                "graph": bool(use_graph),
             },
         }), flush=True)
 
