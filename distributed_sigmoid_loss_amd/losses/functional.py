@@ -111,7 +111,7 @@ def _torch_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
 
 
 def chunk_loss_fwd(zimg, ztxt, t_prime, bias, diag_offset=None,
-                   col_chunk=None, impl="auto", quant="bf16"):
+                   col_chunk=None, impl="auto", quant="bf16", qcache=None):
     """Non-differentiable forward of the block loss (used inside hand-written
     autograd Functions).  Returns a scalar tensor."""
     if impl == "auto":
@@ -119,7 +119,7 @@ def chunk_loss_fwd(zimg, ztxt, t_prime, bias, diag_offset=None,
     if impl == "hip":
         from .. import ops
         return ops.siglip_fwd(zimg.contiguous(), ztxt.contiguous(), t_prime,
-                              bias, diag_offset, quant=quant)
+                              bias, diag_offset, quant=quant, qcache=qcache)
     with torch.no_grad():
         return _torch_loss(zimg, ztxt, t_prime, bias, diag_offset, col_chunk)
 

@@ -82,6 +82,20 @@ class _RingAllGatherLoss(torch.autograd.Function):
         chunks = [None] * world          # indexed by source rank
         chunks[rank] = ztxt
 
+        # fp8: quantize the image shard ONCE for all W chunk losses; each
+        # received text chunk gets its own per-tensor scale.
+        use_q = quant == "fp8" and zimg.is_cuda and impl != "torch"
+        if use_q:
+            from .. import ops as _ops
+            zi_q, si = _ops._quant_fp8(zimg)
+
+            def qc_for(zt_chunk):
+                zt_q, st = _ops._quant_fp8(zt_chunk)
+                return (zi_q, si, zt_q, st)
+        else:
+            def qc_for(zt_chunk):
+                return None
+
         if world > 1:
             left = (rank - 1 + world) % world
             right = (rank + 1) % world
@@ -90,7 +104,8 @@ class _RingAllGatherLoss(torch.autograd.Function):
             handle = neighbour_exchange_start(left, right, ztxt, group=group)
 
         loss = chunk_loss_fwd(zimg, ztxt, t_prime, bias, diag_offset=0,
-                              col_chunk=col_chunk, impl=impl, quant=quant)
+                              col_chunk=col_chunk, impl=impl, quant=quant,
+                              qcache=qc_for(ztxt))
 
         if world > 1:
             for hop in range(1, world):
@@ -104,7 +119,8 @@ class _RingAllGatherLoss(torch.autograd.Function):
                 loss = loss + chunk_loss_fwd(zimg, recv, t_prime, bias,
                                              diag_offset=None,
                                              col_chunk=col_chunk, impl=impl,
-                                             quant=quant)
+                                             quant=quant,
+                                             qcache=qc_for(recv))
 
         ctx.save_for_backward(zimg, t_prime, bias, *chunks)
         ctx.group = group
