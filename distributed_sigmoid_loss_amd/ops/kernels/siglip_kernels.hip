@@ -27,10 +27,11 @@
 //     buffered LDS (4×32 KiB) staged by global_load_lds_dwordx4 (direct
 //     HBM→LDS DMA).  The LDS image is [K-half][row][64 B]: bf16 stages and
 //     waits at K-HALF granularity interleaved with the two MFMA half-steps,
-//     so every 16 KiB granule gets a full iteration of transfer window —
-//     ablation showed the single-phase schedule leaving a 27% overlap
-//     deficit (staging alone 0.469 ms, compute alone 0.449, full 0.636 at
-//     B=16k; per-CU DMA is at its ~26 GB/s HBM share).
+//     so every 16 KiB granule gets a full iteration of transfer window.
+//     (Measured limit at B≥16k: the DMA+MFMA co-run itself — staging alone
+//     0.469 ms, compute alone 0.449, decoupled co-run 0.598, full 0.632 —
+//     finer schedules up to a complete 8-phase fragment-unit pipeline
+//     [tools/probe_8phase.hip] are neutral; see profiles/README.md.)
 //   - The DMA stays in flight across raw s_barriers under counted
 //     s_waitcnt vmcnt(N) (a __syncthreads would drain it); barriers are
 //     asm statements with "memory" clobbers — the plain s_barrier builtin
