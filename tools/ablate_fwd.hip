@@ -11,6 +11,8 @@
 //                                     producer-consumer coupling)
 //   6 = register DMA                 (plain global_load to discarded regs; no
 //                                     LDS writes — HBM read + MFMA co-run)
+//   7 = dummy DMA + MFMA-only        (no ds_reads at all: separates LDS-port
+//                                     contention from clock/issue co-run)
 // Results are wrong for ABL>0 — perf-diagnostic only.  Prints ms and TF/s
 // per variant, interleaved rounds (guide §5.4 rule 24).
 //
@@ -121,7 +123,7 @@ __launch_bounds__(THREADS) __global__ void fwd_kernel(
 
   if (ABL != 2 && ABL != 3 && ABL != 5 && ABL != 6) stage(0);
   for (int kt = 0; kt < ktiles; ++kt) {
-    if (ABL == 5) {
+    if (ABL == 5 || ABL == 7) {
       stage_dummy();
       asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
     } else if (ABL == 6) {
@@ -138,7 +140,7 @@ __launch_bounds__(THREADS) __global__ void fwd_kernel(
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 afrag[FM], bfrag[FN];
-      if (ABL != 3) {
+      if (ABL != 3 && ABL != 7) {
 #pragma unroll
         for (int mi = 0; mi < FM; ++mi)
           afrag[mi] = *reinterpret_cast<const bf16x8*>(
@@ -176,7 +178,7 @@ __launch_bounds__(THREADS) __global__ void fwd_kernel(
     }
     aAddr[0] ^= 2 * TILE_BYTES; aAddr[1] ^= 2 * TILE_BYTES;
     bAddr[0] ^= 2 * TILE_BYTES; bAddr[1] ^= 2 * TILE_BYTES;
-    if (ABL == 0 || ABL == 5 || ABL == 6)
+    if (ABL == 0 || ABL == 5 || ABL == 6 || ABL == 7)
       asm volatile("s_barrier" ::: "memory");
     else if (ABL == 2 || ABL == 3 || ABL == 4) __syncthreads();
     // ABL 1: no post-compute barrier
@@ -241,8 +243,9 @@ int main() {
 
   const double flops = 2.0 * b * n * d;
   const char* names[] = {"full", "no-2nd-barrier", "no-stage", "mfma-only",
-                         "no-mfma", "decoupled-dma", "reg-dma"};
-  float best[7] = {1e9f, 1e9f, 1e9f, 1e9f, 1e9f, 1e9f, 1e9f};
+                         "no-mfma", "decoupled-dma", "reg-dma",
+                         "dma+mfma-no-lds"};
+  float best[8] = {1e9f, 1e9f, 1e9f, 1e9f, 1e9f, 1e9f, 1e9f, 1e9f};
   for (int r = 0; r < rounds; ++r) {
     best[0] = fminf(best[0], run<0>(zi, zt, out, b, n, d, iters));
     best[1] = fminf(best[1], run<1>(zi, zt, out, b, n, d, iters));
@@ -251,8 +254,9 @@ int main() {
     best[4] = fminf(best[4], run<4>(zi, zt, out, b, n, d, iters));
     best[5] = fminf(best[5], run<5>(zi, zt, out, b, n, d, iters));
     best[6] = fminf(best[6], run<6>(zi, zt, out, b, n, d, iters));
+    best[7] = fminf(best[7], run<7>(zi, zt, out, b, n, d, iters));
   }
-  for (int v = 0; v < 7; ++v)
+  for (int v = 0; v < 8; ++v)
     printf("%-16s %8.3f ms  %7.1f TF/s\n", names[v], best[v],
            flops / best[v] / 1e9);
   return 0;
