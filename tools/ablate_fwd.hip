@@ -121,7 +121,7 @@ __launch_bounds__(THREADS) __global__ void fwd_kernel(
     bbase += 128;
   };
 
-  if (ABL != 2 && ABL != 3 && ABL != 5 && ABL != 6) stage(0);
+  if (ABL != 2 && ABL != 3 && ABL != 5 && ABL != 6 && ABL != 7) stage(0);
   for (int kt = 0; kt < ktiles; ++kt) {
     if (ABL == 5 || ABL == 7) {
       stage_dummy();
