@@ -219,11 +219,13 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     if quant == "fp8" and not fp8_ok:
         raise RuntimeError("fp8 backward requires batch % 4 == 0")
 
-    def mm8(a8, b8_rowmajor, s_ab, out_rows):
-        """(out_rows, d) = a8 @ b8 with b8 given row-major (k, d)."""
+    _one = torch.ones((), device=dev)
+
+    def mm8(a8, b8_rowmajor, s_ab):
+        """a8 (m, k) @ b8 (k, d) → bf16; mat2 re-laid column-major as
+        _scaled_mm requires (k ≤ a few thousand rows — cheap copy)."""
         b_cm = b8_rowmajor.t().contiguous().t()   # column-major (k, d)
-        return torch._scaled_mm(a8, b_cm, scale_a=s_ab,
-                                scale_b=torch.ones((), device=dev),
+        return torch._scaled_mm(a8, b_cm, scale_a=s_ab, scale_b=_one,
                                 out_dtype=torch.bfloat16)
 
     def run_g(j0, j1, g_slab, diag, gt_slab=None):
@@ -264,10 +266,10 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         if use_mm8:
             s_t = ((scale / 448.0) * st).reshape(())
             s_i = ((scale / 448.0) * si).reshape(())
-            dztxt = mm8(gt, zi_k, s_i, n)
+            dztxt = mm8(gt, zi_k, s_i)
             if on_dztxt is not None:
                 on_dztxt(dztxt)
-            dzimg = mm8(g, zt_k, s_t, b)
+            dzimg = mm8(g, zt_k, s_t)
         else:
             if quant == "fp8":
                 g = g.to(torch.bfloat16) * (1.0 / 448.0)
@@ -295,7 +297,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
             if use_mm8 and c % 16 == 0:
                 s_t = ((scale / 448.0) * st).reshape(())
                 s_i = ((scale / 448.0) * si).reshape(())
-                dzimg_acc += mm8(g, zt_k[j0:j1], s_t, b).float()
+                dzimg_acc += mm8(g, zt_k[j0:j1], s_t).float()
                 dztxt[j0:j1] = mm8(gt, zi_k, s_i, c)
             else:
                 g16 = (g.to(torch.bfloat16) * (1.0 / 448.0)
