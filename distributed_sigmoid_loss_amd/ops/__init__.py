@@ -298,7 +298,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                 s_t = ((scale / 448.0) * st).reshape(())
                 s_i = ((scale / 448.0) * si).reshape(())
                 dzimg_acc += mm8(g, zt_k[j0:j1], s_t).float()
-                dztxt[j0:j1] = mm8(gt, zi_k, s_i, c)
+                dztxt[j0:j1] = mm8(gt, zi_k, s_i)
             else:
                 g16 = (g.to(torch.bfloat16) * (1.0 / 448.0)
                        ) if quant == "fp8" else g
