@@ -130,3 +130,25 @@ def test_fp8_chunked_matches_single_slab():
     torch.cuda.synchronize()
     for a, b_ in zip(full, chunked):
         assert torch.allclose(a.float(), b_.float(), rtol=3e-2, atol=1e-3)
+
+
+def test_fp8_bwd_fallback_unaligned_shapes():
+    """b%16 != 0 disables the _scaled_mm path → dequantized-g rocBLAS
+    fallback (and the guarded kernel path for the ragged tile)."""
+    b, n, d = 260, 512, 256
+    zi, zt, tp, bs = make_inputs(b, n, d, seed=41)
+    go = torch.tensor(1.0, device="cuda")
+    dzi, dzt, dtp, dbs = ops.siglip_bwd(zi, zt, tp, bs, 100, go, None,
+                                        quant="fp8")
+    zi_q, zt_q = dequant_ref(zi), dequant_ref(zt)
+    r = _torch_bwd(zi_q, zt_q, tp.float(), bs.float(), 100, go.float(),
+                   col_chunk=None)
+    torch.cuda.synchronize()
+
+    def rel_l2(a, b_):
+        return ((a.float() - b_).norm() / b_.norm().clamp(min=1e-12)).item()
+
+    assert rel_l2(dzi, r[0]) < 5e-2
+    assert rel_l2(dzt, r[1]) < 5e-2
+    assert torch.allclose(dtp.float(), r[2], rtol=5e-2, atol=2e-3)
+    assert torch.allclose(dbs.float(), r[3], rtol=5e-2, atol=2e-3)
