@@ -46,8 +46,10 @@ def parse_args():
     p.add_argument("--strategy", choices=["ring", "all_gather"],
                    default="ring")
     p.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
-    p.add_argument("--quant", choices=["bf16", "fp8"], default="bf16",
-                   help="logit-kernel compute dtype (fp8 = MX-scaled e4m3)")
+    p.add_argument("--quant", choices=["bf16", "fp8", "mixed"],
+                   default="bf16",
+                   help="bf16 | fp8 (MX-scaled e4m3 logits + fp8 grad GEMMs)"
+                        " | mixed (bf16 logits, fp8 grad GEMMs)")
     p.add_argument("--impl", choices=["auto", "hip", "torch"], default="auto",
                    help="'torch' = stock-PyTorch floor (materialized logits)")
     p.add_argument("--device", choices=["cuda", "cpu"], default=None)

@@ -158,7 +158,7 @@ class _FusedSigmoidLoss(torch.autograd.Function):
         zimg = zimg.contiguous()
         ztxt = ztxt.contiguous()
         qc = (ops.quantize_fp8_pair(zimg, ztxt)
-              if (quant == "fp8" and zimg.is_cuda) else None)
+              if (quant in ("fp8", "mixed") and zimg.is_cuda) else None)
         loss = ops.siglip_fwd(zimg, ztxt, t_prime, bias, diag_offset,
                               quant=quant, qcache=qc)
         saved = (zimg, ztxt, t_prime, bias) + (qc if qc is not None else ())

@@ -219,7 +219,7 @@ class DistributedSigmoidLoss(nn.Module):
         self.gpu_batch_size = gpu_batch_size
         if strategy not in ("all_gather", "ring"):
             raise ValueError(f"unknown strategy {strategy!r}")
-        if quant not in ("bf16", "fp8"):
+        if quant not in ("bf16", "fp8", "mixed"):
             raise ValueError(f"unknown quant {quant!r}")
         self.strategy = strategy
         self.col_chunk = col_chunk

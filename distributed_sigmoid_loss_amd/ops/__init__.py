@@ -67,6 +67,9 @@ def _load():
     lib.siglip_fwd_fp8.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] * 5
     lib.siglip_bwd_g_fp8.restype = ctypes.c_int
     lib.siglip_bwd_g_fp8.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5
+    lib.siglip_bwd_g_mixed.restype = ctypes.c_int
+    lib.siglip_bwd_g_mixed.argtypes = (
+        [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5)
     _lib = lib
     return _lib
 
@@ -99,7 +102,7 @@ def _validate(zimg: torch.Tensor, ztxt: torch.Tensor, quant: str):
         raise RuntimeError(
             f"siglip HIP ops require bf16 embeddings (got {zimg.dtype}); "
             "cast with .bfloat16() or pass impl='torch'")
-    if quant not in ("bf16", "fp8"):
+    if quant not in ("bf16", "fp8", "mixed"):
         raise ValueError(f"unknown quant {quant!r}")
     mult = 16 if quant == "fp8" else 8
     if zimg.shape[1] % mult != 0:
@@ -135,6 +138,8 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                quant: str = "bf16", qcache=None) -> torch.Tensor:
     lib = _require_lib()
     _validate(zimg, ztxt, quant)
+    if quant == "mixed":
+        quant = "bf16"   # mixed = bf16 logits; fp8 applies to backward only
     b, d = zimg.shape
     n = ztxt.shape[0]
     dev = zimg.device
@@ -179,15 +184,23 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     tp = _prep_scalar(t_prime, dev)
     bp = _prep_scalar(bias, dev)
     t_true = tp.exp()
-    if quant == "fp8":
+    if quant in ("fp8", "mixed"):
         # Reuse the forward's quantization when provided (identical inputs
         # give identical amax, so recomputing is equivalent but wasteful);
         # the g kernel sees t_eff so its logits match the forward's.
         zi_q, si, zt_q, st = (qcache if qcache is not None
                               else quantize_fp8_pair(zimg, ztxt))
-        tp_k = tp + si.log() + st.log()
-        g_fn = lib.siglip_bwd_g_fp8
-        zi_k, zt_k = zi_q, zt_q
+        if quant == "fp8":
+            # fp8 logits: kernel reads quantized inputs with t_eff.
+            tp_k = tp + si.log() + st.log()
+            g_fn = lib.siglip_bwd_g_fp8
+            zi_k, zt_k = zi_q, zt_q
+        else:
+            # mixed: bf16 logits recompute; quantized inputs feed only the
+            # fp8 gradient GEMMs below.
+            tp_k = tp
+            g_fn = lib.siglip_bwd_g_mixed
+            zi_k, zt_k = zimg, ztxt
     else:
         tp_k = tp
         g_fn = lib.siglip_bwd_g_bf16
@@ -198,8 +211,9 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     # Column slab sizing: the kernel's g-store addressing is 32-bit, so
     # b * slab * esz bytes must stay below 2^32; beyond that (and to bound
     # workspace at huge n) we chunk.
-    g_esz = 1 if quant == "fp8" else 2
-    g_dtype = torch.float8_e4m3fn if quant == "fp8" else torch.bfloat16
+    fp8g = quant in ("fp8", "mixed")   # g slabs are e4m3 (×448)
+    g_esz = 1 if fp8g else 2
+    g_dtype = torch.float8_e4m3fn if fp8g else torch.bfloat16
     step = col_chunk if col_chunk and col_chunk > 0 else n
     while (b * step * g_esz) >= 2 ** 32:
         step //= 2
@@ -213,11 +227,10 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     # quantized embeddings carry per-tensor scales — all folded into
     # _scaled_mm's scale args so no extra elementwise passes run.  Shapes
     # must be 16-aligned; otherwise dequantize and use rocBLAS.
-    use_mm8 = (quant == "fp8" and b % 16 == 0 and n % 16 == 0
+    use_mm8 = (fp8g and b % 16 == 0 and n % 16 == 0
                and d % 16 == 0 and hasattr(torch, "_scaled_mm"))
-    fp8_ok = quant == "fp8" and b % 4 == 0   # packed gt stores need b%4
-    if quant == "fp8" and not fp8_ok:
-        raise RuntimeError("fp8 backward requires batch % 4 == 0")
+    if fp8g and b % 4 != 0:   # packed gt stores need b%4
+        raise RuntimeError(f"{quant} backward requires batch % 4 == 0")
 
     _one = torch.ones((), device=dev)
 
@@ -230,7 +243,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
 
     def run_g(j0, j1, g_slab, diag, gt_slab=None):
         zt_slab = zt_k[j0:j1]
-        if quant == "fp8":
+        if fp8g:
             _check(g_fn(
                 ctypes.c_void_p(stream),
                 ctypes.c_void_p(zi_k.data_ptr()),
@@ -241,7 +254,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                 ctypes.c_void_p(gt_slab.data_ptr()),
                 ctypes.c_void_p(scal.data_ptr()),
                 b, j1 - j0, d, diag, _kernel_flags()),
-                "siglip_bwd_g_fp8")
+                "siglip_bwd_g_fp8/mixed")
         else:
             _check(g_fn(
                 ctypes.c_void_p(stream),
@@ -260,7 +273,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         # reduce-scatter) overlaps with the dzimg GEMM below.
         g = torch.empty((b, n), device=dev, dtype=g_dtype)
         gt = (torch.empty((n, b), device=dev, dtype=g_dtype)
-              if quant == "fp8" else None)
+              if fp8g else None)
         run_g(0, n, g, _DIAG_NONE if diag_offset is None else int(diag_offset),
               gt)
         if use_mm8:
@@ -271,7 +284,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                 on_dztxt(dztxt)
             dzimg = mm8(g, zt_k, s_t)
         else:
-            if quant == "fp8":
+            if fp8g:
                 g = g.to(torch.bfloat16) * (1.0 / 448.0)
             dztxt = ((g.T @ zimg) * scale).to(ztxt.dtype)
             if on_dztxt is not None:
@@ -282,14 +295,14 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         dztxt = torch.empty((n, d), device=dev, dtype=torch.bfloat16)
         g_buf = torch.empty((b, step), device=dev, dtype=g_dtype)
         gt_buf = (torch.empty((step, b), device=dev, dtype=g_dtype)
-                  if quant == "fp8" else None)
+                  if fp8g else None)
         for j0 in range(0, n, step):
             j1 = min(j0 + step, n)
             c = j1 - j0
             g = g_buf if c == step else torch.empty(
                 (b, c), device=dev, dtype=g_dtype)
             gt = None
-            if quant == "fp8":
+            if fp8g:
                 gt = gt_buf if c == step else torch.empty(
                     (c, b), device=dev, dtype=g_dtype)
             diag = _DIAG_NONE if diag_offset is None else int(diag_offset) - j0
@@ -301,7 +314,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                 dztxt[j0:j1] = mm8(gt, zi_k, s_i)
             else:
                 g16 = (g.to(torch.bfloat16) * (1.0 / 448.0)
-                       ) if quant == "fp8" else g
+                       ) if fp8g else g
                 # scale applied per chunk so mixed mm8/fallback chunks agree
                 dzimg_acc += ((g16 @ ztxt[j0:j1]) * scale).float()
                 dztxt[j0:j1] = (g16.T @ zimg) * scale

@@ -453,7 +453,7 @@ __device__ __forceinline__ void remap_block(int flags, int& bx, int& by) {
 // Interior-only kernel: every tile full, d % K-step == 0, n%8==0 — checked
 // by the host launcher.  Separate from the general kernel so the hot path's
 // register allocation is not inflated by the guarded path.
-template <int MODE, int EB>
+template <int MODE, int EB, int EB_G = EB>
 __launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
@@ -465,14 +465,14 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
   remap_block(flags, bx, by);
   const float t = __expf(*t_prime);
   const float bias = *bias_p;
-  tile_body<MODE, true, EB>(zimg, ztxt, t, bias, out, g_out, gt_out,
-                            b, n, d, diag, bx * BM, by * BN, smem);
+  tile_body<MODE, true, EB, EB_G>(zimg, ztxt, t, bias, out, g_out, gt_out,
+                                  b, n, d, diag, bx * BM, by * BN, smem);
 }
 
 // General kernel: interior blocks take the DMA fast path, edge blocks the
 // guarded register-staged path.  MODE 0: forward loss.  MODE 1: backward
 // g-slab + scalar partials.
-template <int MODE, int EB>
+template <int MODE, int EB, int EB_G = EB>
 __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
@@ -491,16 +491,17 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
   const bool interior = (row_base + BM <= b) && (col_base + BN <= n) &&
       (d % (128 / EB) == 0) && (n % 8 == 0);
   if (interior)
-    tile_body<MODE, true, EB>(zimg, ztxt, t, bias, out, g_out, gt_out,
-                              b, n, d, diag, row_base, col_base, smem);
+    tile_body<MODE, true, EB, EB_G>(zimg, ztxt, t, bias, out, g_out, gt_out,
+                                    b, n, d, diag, row_base, col_base, smem);
   else
-    tile_body<MODE, false, EB>(zimg, ztxt, t, bias, out, g_out, gt_out,
-                               b, n, d, diag, row_base, col_base, smem);
+    tile_body<MODE, false, EB, EB_G>(zimg, ztxt, t, bias, out, g_out,
+                                     gt_out, b, n, d, diag, row_base,
+                                     col_base, smem);
 }
 
 inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
-template <int MODE, int EB>
+template <int MODE, int EB, int EB_G = EB>
 int launch(uintptr_t stream, const void* zimg, const void* ztxt,
            const void* t_prime, const void* bias, void* out, void* g_out,
            void* gt_out, int b, int n, int d, int diag, int flags) {
@@ -510,14 +511,15 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
   const bool interior = (b % BM == 0) && (n % BN == 0) &&
       (d % (128 / EB) == 0);
   if (interior)
-    hipLaunchKernelGGL((siglip_tile_kernel_interior<MODE, EB>), grid,
+    hipLaunchKernelGGL((siglip_tile_kernel_interior<MODE, EB, EB_G>), grid,
                        dim3(THREADS), 0, (hipStream_t)stream,
                        (const char*)zimg, (const char*)ztxt,
                        (const float*)t_prime, (const float*)bias,
                        (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
                        b, n, d, diag, flags);
   else
-    hipLaunchKernelGGL((siglip_tile_kernel<MODE, EB>), grid, dim3(THREADS),
+    hipLaunchKernelGGL((siglip_tile_kernel<MODE, EB, EB_G>), grid,
+                       dim3(THREADS),
                        0, (hipStream_t)stream,
                        (const char*)zimg, (const char*)ztxt,
                        (const float*)t_prime, (const float*)bias,
@@ -551,6 +553,17 @@ int siglip_fwd_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
                    int b, int n, int d, int diag, int flags) {
   return launch<0, 1>(stream, zimg, ztxt, t_prime, bias, loss_out, nullptr,
                       nullptr, b, n, d, diag, flags);
+}
+
+// Mixed policy: bf16 logits recompute, e4m3 ×448 g and gᵀ slabs (for the
+// fp8 gradient GEMMs) — full-precision loss surface, compressed grads.
+int siglip_bwd_g_mixed(uintptr_t stream, const void* zimg, const void* ztxt,
+                       const void* t_prime, const void* bias, void* g_out,
+                       void* gt_out, void* scal, int b, int n, int d,
+                       int diag, int flags) {
+  if (b % 4 != 0) return (int)hipErrorInvalidValue;
+  return launch<1, 2, 1>(stream, zimg, ztxt, t_prime, bias, scal, g_out,
+                         gt_out, b, n, d, diag, flags);
 }
 
 // fp8 backward emits g (b,n) AND its transpose gt (n,b), both e4m3 ×448.

@@ -152,3 +152,29 @@ def test_fp8_bwd_fallback_unaligned_shapes():
     assert rel_l2(dzt, r[1]) < 5e-2
     assert torch.allclose(dtp.float(), r[2], rtol=5e-2, atol=2e-3)
     assert torch.allclose(dbs.float(), r[3], rtol=5e-2, atol=2e-3)
+
+
+def test_mixed_policy_bwd_matches_bf16():
+    """quant='mixed': bf16 logits with fp8 gradient GEMMs — grads must stay
+    close to the pure-bf16 path (g and GEMM operands quantized e4m3)."""
+    b, n, d = 512, 512, 768
+    zi, zt, tp, bs = make_inputs(b, n, d, seed=55)
+    go = torch.tensor(1.0, device="cuda")
+    mixed = ops.siglip_bwd(zi, zt, tp, bs, 0, go, None, quant="mixed")
+    ref = ops.siglip_bwd(zi, zt, tp, bs, 0, go, None, quant="bf16")
+    l_mixed = ops.siglip_fwd(zi, zt, tp, bs, 0, quant="mixed")
+    l_bf16 = ops.siglip_fwd(zi, zt, tp, bs, 0, quant="bf16")
+    torch.cuda.synchronize()
+    # identical forward (mixed logits ARE bf16)
+    assert torch.allclose(l_mixed, l_bf16, rtol=1e-5, atol=1e-3)
+
+    def rel_l2(a, b_):
+        bf = b_.float()
+        return ((a.float() - bf).norm() / bf.norm().clamp(min=1e-12)).item()
+
+    assert rel_l2(mixed[0], ref[0]) < 3e-2
+    assert rel_l2(mixed[1], ref[1]) < 3e-2
+    assert torch.allclose(mixed[2].float(), ref[2].float(), rtol=2e-2,
+                          atol=1e-3)
+    assert torch.allclose(mixed[3].float(), ref[3].float(), rtol=2e-2,
+                          atol=1e-3)
