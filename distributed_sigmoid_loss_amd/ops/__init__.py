@@ -190,6 +190,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         # the g kernel sees t_eff so its logits match the forward's.
         zi_q, si, zt_q, st = (qcache if qcache is not None
                               else quantize_fp8_pair(zimg, ztxt))
+        zi_g, zt_g = zi_q, zt_q           # fp8 GEMM operands
         if quant == "fp8":
             # fp8 logits: kernel reads quantized inputs with t_eff.
             tp_k = tp + si.log() + st.log()
@@ -279,10 +280,10 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         if use_mm8:
             s_t = ((scale / 448.0) * st).reshape(())
             s_i = ((scale / 448.0) * si).reshape(())
-            dztxt = mm8(gt, zi_k, s_i)
+            dztxt = mm8(gt, zi_g, s_i)
             if on_dztxt is not None:
                 on_dztxt(dztxt)
-            dzimg = mm8(g, zt_k, s_t)
+            dzimg = mm8(g, zt_g, s_t)
         else:
             if fp8g:
                 g = g.to(torch.bfloat16) * (1.0 / 448.0)
@@ -310,8 +311,8 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
             if use_mm8 and c % 16 == 0:
                 s_t = ((scale / 448.0) * st).reshape(())
                 s_i = ((scale / 448.0) * si).reshape(())
-                dzimg_acc += mm8(g, zt_k[j0:j1], s_t).float()
-                dztxt[j0:j1] = mm8(gt, zi_k, s_i)
+                dzimg_acc += mm8(g, zt_g[j0:j1], s_t).float()
+                dztxt[j0:j1] = mm8(gt, zi_g, s_i)
             else:
                 g16 = (g.to(torch.bfloat16) * (1.0 / 448.0)
                        ) if fp8g else g
