@@ -175,7 +175,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": args.dtype if args.quant == "bf16" else "fp8",
+            "dtype": args.dtype if args.quant == "bf16" else args.quant,
             "data": "synthetic",
             "config": {
                 "model": "two-tower-linear-siglip",
