@@ -5,10 +5,15 @@ exposes:
 
 - :func:`siglip_fwd` — fused logits+logsigmoid+sum forward over a (b, n)
   block; the logits matrix never leaves MFMA accumulators.
-- :func:`siglip_bwd` — backward: fused recompute kernel emits the g slab and
-  the scalar partials; the two ``(b,n)×(n,d)`` gradient GEMMs run on rocBLAS
-  via ``torch.matmul`` (plain library GEMMs), column-chunked so workspace is
-  O(b · col_chunk) regardless of n.
+- :func:`siglip_bwd` — backward: fused recompute kernel emits the
+  dL/dlogit slab(s) and the scalar partials; the two ``(b,n)×(n,d)``
+  gradient GEMMs run on the GEMM libraries (rocBLAS bf16 via
+  ``torch.matmul``, or hipBLASLt fp8 via ``torch._scaled_mm`` for the
+  fp8/mixed policies, consuming the kernel-emitted e4m3 g and gᵀ slabs),
+  column-chunked so workspace is O(b · col_chunk) regardless of n.
+- quantization policies: ``bf16`` | ``fp8`` (e4m3 logits via the MX-scaled
+  MFMA, per-tensor scales folded into the temperature) | ``mixed`` (bf16
+  logits, fp8 gradient GEMMs).
 
 These are *loud* paths: calling them on a GPU without the built extension
 raises — there is no silent eager fallback on device (CPU fallbacks live in
@@ -172,9 +177,12 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                quant: str = "bf16", on_dztxt=None, qcache=None):
     """Returns (dzimg, dztxt, dt_prime, dbias).
 
-    Per column slab: the fused kernel recomputes logit tiles (MFMA) and writes
-    g = dL/d(logit pre-scale); then dzimg += g @ ztxt_slab and
-    dztxt_slab = gᵀ @ zimg (rocBLAS bf16 GEMMs, fp32 accumulation buffers).
+    Per column slab: the fused kernel recomputes logit tiles (MFMA) and
+    writes g = dL/d(logit pre-scale); then dzimg += g @ ztxt_slab and
+    dztxt_slab = gᵀ @ zimg on the GEMM libraries (bf16 rocBLAS, or fp8
+    hipBLASLt when the policy provides e4m3 slabs).  ``on_dztxt(dztxt)`` is
+    invoked as soon as the text gradient exists so distributed callers can
+    overlap their reduce-scatter with the remaining image-gradient GEMM.
     """
     lib = _require_lib()
     _validate(zimg, ztxt, quant)

@@ -4,7 +4,8 @@
 // dim contiguous; bf16 or OCP fp8-e4m3):
 //
 //   forward:  loss += sum_ij softplus(-l_ij * (t * <zimg_i, ztxt_j> + bias))
-//   backward: g_ij  = -l_ij * sigmoid(-l_ij * z_ij)   (written as bf16 slab)
+//   backward: g_ij  = -l_ij * sigmoid(-l_ij * z_ij)   (slab: bf16, or e4m3
+//             ×448 plus its transpose for the fp8/mixed grad-GEMM policies)
 //             scal[0] += sum g_ij * <zimg_i, ztxt_j>   (for dt')
 //             scal[1] += sum g_ij                      (for dbias)
 //
