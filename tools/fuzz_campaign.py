@@ -1,0 +1,91 @@
+#!/usr/bin/env python3
+"""One-off deep fuzz: many random shapes, fwd+bwd parity vs fp32 torch.
+
+Broader than the seeded suite in tests/test_gpu_fuzz.py; run manually on a
+GPU box:  python tools/fuzz_campaign.py [n_cases] [seed]
+"""
+
+import math
+import os
+import random
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+import torch.nn.functional as F
+
+from distributed_sigmoid_loss_amd import ops
+from distributed_sigmoid_loss_amd.losses.functional import (
+    _torch_loss,
+    _torch_bwd,
+)
+
+
+def rel_l2(a, b):
+    b = b.float()
+    return ((a.float() - b).norm() / b.norm().clamp(min=1e-12)).item()
+
+
+def main():
+    n_cases = int(sys.argv[1]) if len(sys.argv) > 1 else 100
+    seed = int(sys.argv[2]) if len(sys.argv) > 2 else 20260913
+    rng = random.Random(seed)
+    failures = 0
+    for i in range(n_cases):
+        quant = "fp8" if i % 3 == 2 else "bf16"
+        b = rng.randrange(1, 1400)
+        n = rng.randrange(1, 1400)
+        dm = 16 if quant == "fp8" else 8
+        d = dm * rng.randrange(1, 112 * 8 // dm + 1)
+        if quant == "fp8":
+            b = max(4, (b // 4) * 4)
+        diag = rng.choice([None, 0, rng.randrange(-b, n + 1)])
+        g = torch.Generator().manual_seed(rng.randrange(1 << 30))
+        zi = F.normalize(torch.randn(b, d, generator=g), dim=-1) \
+            .cuda().bfloat16()
+        zt = F.normalize(torch.randn(n, d, generator=g), dim=-1) \
+            .cuda().bfloat16()
+        tp = torch.tensor(rng.uniform(0.0, 3.0), device="cuda")
+        bs = torch.tensor(rng.uniform(-12.0, 0.0), device="cuda")
+        go = torch.tensor(rng.uniform(0.2, 2.0), device="cuda")
+        col = rng.choice([None, 256 * rng.randrange(1, 5)])
+        try:
+            got = ops.siglip_fwd(zi, zt, tp, bs, diag, quant=quant)
+            dk = ops.siglip_bwd(zi, zt, tp, bs, diag, go, col, quant=quant)
+            if quant == "fp8":
+                qz = ops._quant_fp8(zi)
+                qt = ops._quant_fp8(zt)
+                ri = qz[0].float() * qz[1]
+                rt = qt[0].float() * qt[1]
+            else:
+                ri, rt = zi.float(), zt.float()
+            want = _torch_loss(ri, rt, tp.float(), bs.float(), diag, None)
+            dr = _torch_bwd(ri, rt, tp.float(), bs.float(), diag, go.float(),
+                            None)
+            torch.cuda.synchronize()
+            ok = True
+            fl = abs(got.item() - want.item()) / max(abs(want.item()), 1e-6)
+            if fl > 3e-2:
+                ok = False
+            errs = [rel_l2(dk[0], dr[0]), rel_l2(dk[1], dr[1]),
+                    rel_l2(dk[2], dr[2]), rel_l2(dk[3], dr[3])]
+            # fp8 grad GEMMs use original-bf16/quantized mix → looser
+            tol = 8e-2 if quant == "fp8" else 3e-2
+            if max(errs) > tol:
+                ok = False
+            if not ok:
+                failures += 1
+                print(f"FAIL case {i}: q={quant} b={b} n={n} d={d} "
+                      f"diag={diag} col={col} floss={fl:.3e} errs="
+                      f"{['%.3e' % e for e in errs]}")
+        except Exception as e:
+            failures += 1
+            print(f"ERROR case {i}: q={quant} b={b} n={n} d={d} diag={diag} "
+                  f"col={col}: {type(e).__name__}: {str(e)[:160]}")
+    print(f"{n_cases - failures}/{n_cases} passed")
+    sys.exit(1 if failures else 0)
+
+
+if __name__ == "__main__":
+    main()
