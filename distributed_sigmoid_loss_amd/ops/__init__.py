@@ -43,6 +43,8 @@ def _kernel_flags() -> int:
         f |= 1
     if os.environ.get("SIGLIP_GROUP_SWZ", "1") != "0":
         f |= 2
+    gm = os.environ.get("SIGLIP_GROUP_M", "8")
+    f |= {"8": 0, "1": 1, "4": 2, "16": 3}.get(gm, 0) << 4
     return f
 
 _lib = None

@@ -440,10 +440,15 @@ __device__ __forceinline__ void remap_block(int flags, int& bx, int& by) {
     id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
   if (flags & 2) {
-    const int group = id / (GROUP_M * gy);
-    const int within = id % (GROUP_M * gy);
-    const int gm = min(GROUP_M, gx - group * GROUP_M);
-    bx = group * GROUP_M + within % gm;
+    // Group height selectable via flags bits 4-5 (00→8, 01→1, 10→4, 11→16)
+    // for locality A/B experiments; 1 = band walk with the A panel
+    // L2-resident for a whole XCD span.
+    const int sel = (flags >> 4) & 3;
+    const int gmh = sel == 0 ? GROUP_M : (sel == 1 ? 1 : (sel == 2 ? 4 : 16));
+    const int group = id / (gmh * gy);
+    const int within = id % (gmh * gy);
+    const int gm = min(gmh, gx - group * gmh);
+    bx = group * gmh + within % gm;
     by = within / gm;
   } else {
     bx = id % gx;
