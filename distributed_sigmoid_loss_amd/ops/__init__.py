@@ -43,7 +43,7 @@ def _kernel_flags() -> int:
         f |= 1
     if os.environ.get("SIGLIP_GROUP_SWZ", "1") != "0":
         f |= 2
-    gm = os.environ.get("SIGLIP_GROUP_M", "8")
+    gm = os.environ.get("SIGLIP_GROUP_M", "4")
     f |= {"8": 0, "1": 1, "4": 2, "16": 3}.get(gm, 0) << 4
     return f
 
