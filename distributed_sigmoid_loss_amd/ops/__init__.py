@@ -34,8 +34,9 @@ _DIAG_NONE = -(2 ** 31)
 
 
 def _kernel_flags() -> int:
-    # bit 0: XCD-contiguous block remap (default off — measured slightly
-    #        negative in the L3-resident regime; SIGLIP_XCD_SWZ=1 enables).
+    # bit 0: XCD-contiguous block remap — default ON: together with the
+    #        grouped walk it measured best (+4%, profiles/flag_matrix_gm4.log
+    #        and profiles/README.md); SIGLIP_XCD_SWZ=0 disables.
     # bit 1: grouped block walk for L2 panel reuse (default on;
     #        SIGLIP_GROUP_SWZ=0 disables for A/B profiling).
     f = 0
@@ -226,9 +227,16 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     g_esz = 1 if fp8g else 2
     g_dtype = torch.float8_e4m3fn if fp8g else torch.bfloat16
     step = col_chunk if col_chunk and col_chunk > 0 else n
-    while (b * step * g_esz) >= 2 ** 32:
+    while step > 256 and (b * step * g_esz) >= 2 ** 32:
         step //= 2
+    # Floor of 256 columns per slab: below that the kernel grid degenerates.
+    # The 32-bit invariant must survive the floor — at bf16 that means
+    # b < 2^32/(256·2) = 8.4M rows, far beyond any real shard.
     step = max(step, 256)
+    if (b * step * g_esz) >= 2 ** 32:
+        raise RuntimeError(
+            f"batch {b} too large for the 32-bit g-slab addressing even at "
+            f"the 256-column slab floor; shard the batch")
 
     go = grad_output.detach().reshape(()).to(device=dev, dtype=torch.float32)
     scale = go * t_true   # gradient GEMMs run against the original bf16

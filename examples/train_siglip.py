@@ -35,11 +35,15 @@ from distributed_sigmoid_loss_amd.utils import init_from_env, set_seed
 from distributed_sigmoid_loss_amd.utils.profiling import PhaseTimer
 
 
-def synthetic_batch(b, dim, device, dtype, step):
+def synthetic_batch(b, dim, device, dtype, seed):
     # Device-side generation — a host dataloader would overlap H2D copies
     # with compute via a prefetching pipeline; synthetic data skips that.
-    img = torch.randn(b, dim, device=device, dtype=dtype)
-    txt = torch.randn(b, dim, device=device, dtype=dtype)
+    # Seeded per (step, rank) so each rank's shard is distinct — with a
+    # shared generator every rank would draw identical batches and the
+    # cross-rank negatives would duplicate the positives.
+    g = torch.Generator(device=device).manual_seed(seed)
+    img = torch.randn(b, dim, device=device, dtype=dtype, generator=g)
+    txt = torch.randn(b, dim, device=device, dtype=dtype, generator=g)
     return img, txt
 
 

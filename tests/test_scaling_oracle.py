@@ -47,6 +47,18 @@ def test_n_rank_matches_single_rank(strategy, world, batch, emb_dim):
             f"{key} grads diverge: {multi[key]} vs {single[key]}"
 
 
+@pytest.mark.parametrize("strategy", ["all_gather", "ring"])
+@pytest.mark.parametrize("emb_dim", [128, 512])
+def test_n_rank_matches_single_rank_wide_dims(strategy, emb_dim):
+    """Reference's wide-dim sweep (test_distributed_sigmoid_loss.py:144-148
+    runs d up to 512) at W=2, batch 4."""
+    multi = run_distributed(ddp_step, 2, 4, emb_dim, strategy, True)[0]
+    single = run_distributed(ddp_step, 1, 8, emb_dim, strategy, True)[0]
+    for key in ("img", "txt", "t_prime", "bias"):
+        assert torch.allclose(multi[key], single[key], rtol=1e-3, atol=1e-6), \
+            f"{key} grads diverge at d={emb_dim}"
+
+
 @pytest.mark.parametrize("world", [2, 3])
 def test_ring_matches_all_gather_raw_grads(world):
     """Both strategies of DistributedSigmoidLoss must be numerically
