@@ -149,20 +149,49 @@ def chunk_loss_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
 
 
 class _FusedSigmoidLoss(torch.autograd.Function):
-    """GPU path: hand-written HIP kernels; backward recomputes logits."""
+    """GPU path: hand-written HIP kernels.
+
+    Two backward regimes (`ops.save_g_enabled` picks):
+
+    - **saved-g** (default when the slab fits — 2 GiB at the B=32k bf16
+      headline config): forward runs the fwd+g kernel emitting loss, the
+      dL/dlogit slab and both scalar partials in one pass; backward is then
+      two GEMMs.  The logits GEMM runs once per step, like the reference's
+      autograd (which saves the logits graph) but at O(b·n) g-slab memory
+      instead of O(b·n) fp32 logits + labels.
+    - **recompute** (huge-batch chunked path, or SIGLIP_SAVE_G=0): forward
+      emits only the scalar; backward recomputes logit tiles slab by slab.
+    """
 
     @staticmethod
     def forward(ctx, zimg, ztxt, t_prime, bias, diag_offset, col_chunk,
-                quant):
+                quant, want_grad):
         from .. import ops
         zimg = zimg.contiguous()
         ztxt = ztxt.contiguous()
         qc = (ops.quantize_fp8_pair(zimg, ztxt)
               if (quant in ("fp8", "mixed") and zimg.is_cuda) else None)
-        loss = ops.siglip_fwd(zimg, ztxt, t_prime, bias, diag_offset,
-                              quant=quant, qcache=qc)
-        saved = (zimg, ztxt, t_prime, bias) + (qc if qc is not None else ())
+        b, n = zimg.shape[0], ztxt.shape[0]
+        # want_grad is computed by the CALLER (grad mode is disabled inside
+        # Function.forward, so it cannot be probed here).
+        save_g = (want_grad and col_chunk is None and zimg.is_cuda
+                  and ops.save_g_enabled(b, n, quant))
+        if save_g:
+            out3, g, gt = ops.siglip_fwd_g(zimg, ztxt, t_prime, bias,
+                                           diag_offset, quant=quant,
+                                           qcache=qc)
+            loss = out3[0].clone()
+            saved = (zimg, ztxt, t_prime, bias, out3, g) \
+                + ((gt,) if gt is not None else ()) \
+                + (qc if qc is not None else ())
+        else:
+            loss = ops.siglip_fwd(zimg, ztxt, t_prime, bias, diag_offset,
+                                  quant=quant, qcache=qc)
+            saved = (zimg, ztxt, t_prime, bias) \
+                + (qc if qc is not None else ())
         ctx.save_for_backward(*saved)
+        ctx.saved_g = save_g
+        ctx.has_gt = save_g and quant in ("fp8", "mixed")
         ctx.diag_offset = diag_offset
         ctx.col_chunk = col_chunk   # None → single slab when addressable
         ctx.quant = quant
@@ -172,11 +201,24 @@ class _FusedSigmoidLoss(torch.autograd.Function):
     def backward(ctx, grad_output):
         from .. import ops
         zimg, ztxt, t_prime, bias = ctx.saved_tensors[:4]
-        qc = ctx.saved_tensors[4:] if len(ctx.saved_tensors) > 4 else None
-        dzimg, dztxt, dt_prime, dbias = ops.siglip_bwd(
-            zimg, ztxt, t_prime, bias, ctx.diag_offset, grad_output,
-            ctx.col_chunk, quant=ctx.quant, qcache=qc)
-        return dzimg, dztxt, dt_prime, dbias, None, None, None
+        if ctx.saved_g:
+            i = 4
+            out3, g = ctx.saved_tensors[i:i + 2]
+            i += 2
+            gt = None
+            if ctx.has_gt:
+                gt = ctx.saved_tensors[i]
+                i += 1
+            qc = ctx.saved_tensors[i:] or None
+            dzimg, dztxt, dt_prime, dbias = ops.siglip_bwd_from_g(
+                zimg, ztxt, t_prime, bias, grad_output, out3, g, gt,
+                quant=ctx.quant, qcache=qc)
+        else:
+            qc = ctx.saved_tensors[4:] if len(ctx.saved_tensors) > 4 else None
+            dzimg, dztxt, dt_prime, dbias = ops.siglip_bwd(
+                zimg, ztxt, t_prime, bias, ctx.diag_offset, grad_output,
+                ctx.col_chunk, quant=ctx.quant, qcache=qc)
+        return dzimg, dztxt, dt_prime, dbias, None, None, None, None
 
 
 def sigmoid_contrastive_loss(zimg: torch.Tensor, ztxt: torch.Tensor,
@@ -207,8 +249,10 @@ def sigmoid_contrastive_loss(zimg: torch.Tensor, ztxt: torch.Tensor,
     if impl == "auto":
         impl = "hip" if zimg.is_cuda else "torch"
     if impl == "hip":
+        want_grad = torch.is_grad_enabled() and any(
+            t.requires_grad for t in (zimg, ztxt, t_prime, bias))
         return _FusedSigmoidLoss.apply(zimg, ztxt, t_prime, bias, diag_offset,
-                                       col_chunk, quant)
+                                       col_chunk, quant, want_grad)
     if impl == "torch":
         return _torch_loss(zimg, ztxt, t_prime, bias, diag_offset, col_chunk)
     raise ValueError(f"unknown impl {impl!r}")

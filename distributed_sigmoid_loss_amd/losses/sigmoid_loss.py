@@ -74,20 +74,35 @@ class _RingAllGatherLoss(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, zimg, ztxt, t_prime, bias, group, col_chunk, impl,
-                quant="bf16"):
+                quant, want_grad):
+        from .. import ops as _ops
         world, rank = _world_and_rank(group)
         zimg = zimg.contiguous()
         ztxt = ztxt.contiguous()
+        b_img = zimg.shape[0]
         b_txt = ztxt.shape[0]
 
         chunks = [None] * world          # indexed by source rank
         chunks[rank] = ztxt
 
+        on_gpu = zimg.is_cuda and impl != "torch"
+        # saved-g: each chunk's fwd+g kernel writes its dL/dlogit block
+        # straight into one (b, W·b) slab (column offset = source rank), so
+        # backward is pure GEMMs — no logits recompute anywhere in the step.
+        # want_grad comes from the caller (grad mode is off inside
+        # Function.forward).
+        save_g = (on_gpu and col_chunk is None and want_grad
+                  and _ops.extension_available()
+                  and _ops.save_g_enabled(b_img, world * b_txt, quant))
+
         # fp8: quantize the image shard ONCE for all W chunk losses; each
-        # received text chunk gets its own per-tensor scale.
-        use_q = quant == "fp8" and zimg.is_cuda and impl != "torch"
+        # received text chunk gets its own per-tensor scale, SAVED for
+        # backward so fwd and bwd see identical quantized values (the
+        # round-1 version requantized the concatenated block with one
+        # scale — fwd/bwd loss surfaces differed at W>1).
+        use_q = quant == "fp8" and on_gpu
+        qcaches = [None] * world        # per-source (zi_q, si, zt_q, st)
         if use_q:
-            from .. import ops as _ops
             zi_q, si = _ops._quant_fp8(zimg)
 
             def qc_for(zt_chunk):
@@ -97,6 +112,39 @@ class _RingAllGatherLoss(torch.autograd.Function):
             def qc_for(zt_chunk):
                 return None
 
+        g_slab = gt_slab = out3 = None
+        g_chunks = [None] * world       # fp8: per-chunk g (scales differ)
+        out3s = [None] * world
+        if save_g:
+            esz = 1 if quant in ("fp8", "mixed") else 2
+            g_dtype = (torch.float8_e4m3fn if esz == 1 else torch.bfloat16)
+            if quant != "fp8":
+                g_slab = torch.empty((b_img, world * b_txt),
+                                     device=zimg.device, dtype=g_dtype)
+                out3 = torch.zeros(3, device=zimg.device, dtype=torch.float32)
+            if quant in ("fp8", "mixed"):
+                gt_slab = torch.empty((world * b_txt, b_img),
+                                      device=zimg.device, dtype=g_dtype)
+
+        def chunk_fwd(zt_chunk, src, diag):
+            qc = qc_for(zt_chunk)
+            qcaches[src] = qc
+            if not save_g:
+                return chunk_loss_fwd(zimg, zt_chunk, t_prime, bias,
+                                      diag_offset=diag, col_chunk=col_chunk,
+                                      impl=impl, quant=quant, qcache=qc)
+            if quant == "fp8":
+                o3, g_c, _ = _ops.siglip_fwd_g(
+                    zimg, zt_chunk, t_prime, bias, diag, quant=quant,
+                    qcache=qc, gt_slab=gt_slab, col0=src * b_txt)
+                g_chunks[src] = g_c
+                out3s[src] = o3
+                return o3[0]
+            o3, _, _ = _ops.siglip_fwd_g(
+                zimg, zt_chunk, t_prime, bias, diag, quant=quant,
+                g_slab=g_slab, gt_slab=gt_slab, col0=src * b_txt, out3=out3)
+            return o3[0]
+
         if world > 1:
             left = (rank - 1 + world) % world
             right = (rank + 1) % world
@@ -104,9 +152,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
             # local-block kernel below.
             handle = neighbour_exchange_start(left, right, ztxt, group=group)
 
-        loss = chunk_loss_fwd(zimg, ztxt, t_prime, bias, diag_offset=0,
-                              col_chunk=col_chunk, impl=impl, quant=quant,
-                              qcache=qc_for(ztxt))
+        loss = chunk_fwd(ztxt, rank, 0)
 
         if world > 1:
             for hop in range(1, world):
@@ -121,13 +167,40 @@ class _RingAllGatherLoss(torch.autograd.Function):
                     handle = neighbour_exchange_start(left, right, recv,
                                                       group=group)
                 with roctx_range(f"ring_chunk{hop}_loss"):
-                    loss = loss + chunk_loss_fwd(zimg, recv, t_prime, bias,
-                                                 diag_offset=None,
-                                                 col_chunk=col_chunk,
-                                                 impl=impl, quant=quant,
-                                                 qcache=qc_for(recv))
+                    loss = loss + chunk_fwd(recv, src, None)
 
-        ctx.save_for_backward(zimg, t_prime, bias, *chunks)
+        if save_g:
+            loss = loss.clone()   # detach from the shared out3 views
+
+        saved = [zimg, t_prime, bias] + chunks
+        idx = {}
+        extra = []
+        if save_g:
+            if quant == "fp8":
+                idx["g_chunks"] = len(extra)
+                extra += g_chunks
+                idx["out3s"] = len(extra)
+                extra += out3s
+                idx["gt_slab"] = len(extra)
+                extra.append(gt_slab)
+            else:
+                idx["g_slab"] = len(extra)
+                extra.append(g_slab)
+                idx["out3"] = len(extra)
+                extra.append(out3)
+                if gt_slab is not None:
+                    idx["gt_slab"] = len(extra)
+                    extra.append(gt_slab)
+        if use_q:
+            # per-source (zt_q, st) + the shared (zi_q, si)
+            idx["zi_q"] = len(extra)
+            extra += [zi_q, si]
+            idx["ztq"] = len(extra)
+            for qc in qcaches:
+                extra += [qc[2], qc[3]]
+        ctx.save_for_backward(*saved, *extra)
+        ctx.extra_idx = idx
+        ctx.save_g = save_g
         ctx.group = group
         ctx.world = world
         ctx.rank = rank
@@ -139,20 +212,20 @@ class _RingAllGatherLoss(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_output):
-        zimg, t_prime, bias = ctx.saved_tensors[:3]
-        chunks = ctx.saved_tensors[3:]
+        from .. import ops as _ops
         world, rank = ctx.world, ctx.rank
         b = ctx.b_txt
+        zimg, t_prime, bias = ctx.saved_tensors[:3]
+        chunks = ctx.saved_tensors[3:3 + world]
+        extra = ctx.saved_tensors[3 + world:]
+        idx = ctx.extra_idx
+        quant = ctx.quant
 
-        # One fused backward over the concatenated (W·b, d) text block —
-        # identical math to per-chunk calls (diag at the own-rank block),
-        # but a single dzimg accumulator and one slab loop.  The cross-rank
-        # reduce-scatter (the collapse of the reference's W−1 reversed ring
-        # hops, distributed_utils.py:74-77, into one RCCL collective) is
-        # launched ASYNC from the on_dztxt hook, so its xGMI wire time
-        # overlaps the dzimg gradient GEMM still running on the compute
-        # stream.
-        all_txt = chunks[0] if world == 1 else torch.cat(chunks, dim=0)
+        # The cross-rank reduce-scatter (the collapse of the reference's
+        # W−1 reversed ring hops, distributed_utils.py:74-77, into one RCCL
+        # collective) is launched ASYNC from the on_dztxt hook, so its xGMI
+        # wire time overlaps the dzimg gradient GEMM(s) still running on
+        # the compute stream.
         comm = {}
 
         def on_dztxt(flat):
@@ -170,10 +243,85 @@ class _RingAllGatherLoss(torch.autograd.Function):
                     async_op=True)
                 comm["out"] = out
 
-        dzimg, dtxt_flat, dt_prime, dbias = chunk_loss_bwd(
-            zimg, all_txt, t_prime, bias, rank * b, grad_output,
-            col_chunk=ctx.col_chunk, impl=ctx.impl, quant=ctx.quant,
-            on_dztxt=on_dztxt)
+        go = grad_output
+        if ctx.save_g and quant != "fp8":
+            # One slab: dztxt_flat = t·go·(gᵀ @ zimg) first (feeds the async
+            # reduce-scatter), then dzimg = t·go·(g @ concat-text).
+            all_txt = chunks[0] if world == 1 else torch.cat(chunks, dim=0)
+            out3 = extra[idx["out3"]]
+            g = extra[idx["g_slab"]]
+            gt = extra[idx["gt_slab"]] if "gt_slab" in idx else None
+            qc = None
+            if quant == "mixed":
+                # mixed: g came from exact bf16 logits; quantization only
+                # compresses the grad GEMM operands — one pass here.
+                qc = _ops.quantize_fp8_pair(zimg, all_txt)
+            dzimg, dtxt_flat, dt_prime, dbias = _ops.siglip_bwd_from_g(
+                zimg, all_txt, t_prime, bias, go, out3, g, gt,
+                quant=quant, qcache=qc, on_dztxt=on_dztxt)
+        elif ctx.save_g:
+            # fp8 saved-g: per-chunk g and scales (text side); the image
+            # side has one scale, so dztxt is still a single GEMM over the
+            # full gt slab.
+            zi_q, si = extra[idx["zi_q"]], extra[idx["zi_q"] + 1]
+            gt_slab = extra[idx["gt_slab"]]
+            tp32 = t_prime.detach().reshape(()).float()
+            t_true = tp32.exp()
+            go32 = go.detach().reshape(()).to(zimg.device).float()
+            scale = go32 * t_true
+            dtxt_flat = _ops.scaled_mm8(gt_slab, zi_q, (scale / 448.0) * si)
+            on_dztxt(dtxt_flat)
+            dzimg_acc = torch.zeros_like(zimg, dtype=torch.float32)
+            dt_acc = torch.zeros((), device=zimg.device, dtype=torch.float32)
+            db_acc = torch.zeros((), device=zimg.device, dtype=torch.float32)
+            for src in range(world):
+                zt_q = extra[idx["ztq"] + 2 * src]
+                st = extra[idx["ztq"] + 2 * src + 1]
+                g_c = extra[idx["g_chunks"] + src]
+                o3 = extra[idx["out3s"] + src]
+                dzimg_acc += _ops.scaled_mm8(
+                    g_c, zt_q, (scale / 448.0) * st).float()
+                dt_acc = dt_acc + o3[1] * (t_true * si * st)
+                db_acc = db_acc + o3[2]
+            dzimg = dzimg_acc.to(zimg.dtype)
+            dt_prime = (dt_acc * go32).to(t_prime.dtype).reshape(
+                t_prime.shape)
+            dbias = (db_acc * go32).to(bias.dtype).reshape(bias.shape)
+        elif quant == "fp8" and "ztq" in idx:
+            # fp8 recompute path: per-chunk backward with the forward's own
+            # per-chunk scales (never requantize — fwd/bwd must see the same
+            # quantized values).
+            zi_q, si = extra[idx["zi_q"]], extra[idx["zi_q"] + 1]
+            dzimg_acc = torch.zeros_like(zimg, dtype=torch.float32)
+            dtxt_flat = torch.empty((world * b, zimg.shape[1]),
+                                    device=zimg.device, dtype=zimg.dtype)
+            dt_prime = None
+            dbias = None
+            for src in range(world):
+                zt_q = extra[idx["ztq"] + 2 * src]
+                st = extra[idx["ztq"] + 2 * src + 1]
+                qc = (zi_q, si, zt_q, st)
+                diag = 0 if src == rank else None
+                dzi, dzt, dtp, dbs = chunk_loss_bwd(
+                    zimg, chunks[src], t_prime, bias, diag, go,
+                    col_chunk=ctx.col_chunk, impl=ctx.impl, quant=quant,
+                    qcache=qc)
+                dzimg_acc += dzi.float()
+                dtxt_flat[src * b:(src + 1) * b] = dzt
+                dt_prime = dtp if dt_prime is None else dt_prime + dtp
+                dbias = dbs if dbias is None else dbias + dbs
+            on_dztxt(dtxt_flat)
+            dzimg = dzimg_acc.to(zimg.dtype)
+        else:
+            # bf16/mixed recompute: one fused backward over the concatenated
+            # (W·b, d) text block — identical math to per-chunk calls (diag
+            # at the own-rank block), a single dzimg accumulator and one
+            # slab loop.
+            all_txt = chunks[0] if world == 1 else torch.cat(chunks, dim=0)
+            dzimg, dtxt_flat, dt_prime, dbias = chunk_loss_bwd(
+                zimg, all_txt, t_prime, bias, rank * b, go,
+                col_chunk=ctx.col_chunk, impl=ctx.impl, quant=quant,
+                on_dztxt=on_dztxt)
 
         if world > 1:
             comm["work"].wait()
@@ -184,7 +332,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
         else:
             dztxt = dtxt_flat
 
-        return dzimg, dztxt, dt_prime, dbias, None, None, None, None
+        return (dzimg, dztxt, dt_prime, dbias, None, None, None, None, None)
 
 
 class DistributedSigmoidLoss(nn.Module):
@@ -237,9 +385,12 @@ class DistributedSigmoidLoss(nn.Module):
                 diag_offset=rank * b_txt, col_chunk=self.col_chunk,
                 impl=self.impl, quant=self.quant)
         else:
+            want_grad = torch.is_grad_enabled() and any(
+                t.requires_grad for t in (image_embeddings, text_embeddings,
+                                          self.t_prime, self.bias))
             total = _RingAllGatherLoss.apply(
                 image_embeddings, text_embeddings, self.t_prime, self.bias,
-                group, self.col_chunk, self.impl, self.quant)
+                group, self.col_chunk, self.impl, self.quant, want_grad)
         return total / self.gpu_batch_size
 
 

@@ -78,6 +78,18 @@ def _load():
     lib.siglip_bwd_g_mixed.restype = ctypes.c_int
     lib.siglip_bwd_g_mixed.argtypes = (
         [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5)
+    lib.siglip_fwdg_bf16.restype = ctypes.c_int
+    lib.siglip_fwdg_bf16.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] * 6
+    lib.siglip_fwdg_mixed.restype = ctypes.c_int
+    lib.siglip_fwdg_mixed.argtypes = (
+        [ctypes.c_void_p] * 8 + [ctypes.c_int] * 6)
+    lib.siglip_fwdg_fp8.restype = ctypes.c_int
+    lib.siglip_fwdg_fp8.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 6
+    if lib.siglip_ext_abi() != 6:
+        _lib_err = (f"stale HIP extension at {SO_PATH} "
+                    f"(ABI {lib.siglip_ext_abi()}, need 6); rebuild with: "
+                    "python -m distributed_sigmoid_loss_amd.ops.build --force")
+        return None
     _lib = lib
     return _lib
 
@@ -348,3 +360,160 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     dt_prime = (scal[0] * go * t_eff).to(t_prime.dtype).reshape(t_prime.shape)
     dbias = (scal[1] * go).to(bias.dtype).reshape(bias.shape)
     return dzimg, dztxt, dt_prime, dbias
+
+
+# ---------------------------------------------------------------------------
+# fwd+g ("saved-g") path: one kernel emits loss + g slab + scalar partials,
+# so backward is pure GEMMs — the training step never computes the logits
+# GEMM twice.  Used whenever the slab fits comfortably in HBM (it is 2 GiB
+# at the B=32k bf16 headline config, against 288 GB); the recompute kernels
+# above remain for the chunked huge-batch path (BASELINE config 3).
+# ---------------------------------------------------------------------------
+
+
+def save_g_enabled(b: int, n: int, quant: str) -> bool:
+    """Policy: is the fwd+g saved-slab path usable/worth it for this block?
+
+    SIGLIP_SAVE_G = auto (default) | 0 (always recompute) | 1 (force when
+    addressable); SIGLIP_SAVE_G_MAX_BYTES bounds the slab workspace in auto
+    mode (default 6 GiB — g plus, for fp8/mixed, its transpose).
+    """
+    env = os.environ.get("SIGLIP_SAVE_G", "auto")
+    if env == "0":
+        return False
+    esz = 1 if quant in ("fp8", "mixed") else 2
+    if b * n * esz >= 2 ** 32:        # kernel's 32-bit slab addressing
+        return False
+    if env == "1":
+        return True
+    need = b * n * esz * (2 if esz == 1 else 1)
+    cap = int(os.environ.get("SIGLIP_SAVE_G_MAX_BYTES", str(6 * 2 ** 30)))
+    return need <= cap
+
+
+def scaled_mm8(a8: torch.Tensor, b8_rowmajor: torch.Tensor,
+               scale: torch.Tensor) -> torch.Tensor:
+    """fp8 GEMM a8 (m, k) @ b8 (k, d) → bf16 via hipBLASLt
+    (``torch._scaled_mm``; mat2 re-laid column-major as it requires — k is at
+    most a few thousand rows, a cheap copy)."""
+    one = torch.ones((), device=a8.device)
+    b_cm = b8_rowmajor.t().contiguous().t()
+    return torch._scaled_mm(a8, b_cm, scale_a=scale.reshape(()), scale_b=one,
+                            out_dtype=torch.bfloat16)
+
+
+def siglip_fwd_g(zimg: torch.Tensor, ztxt: torch.Tensor,
+                 t_prime: torch.Tensor, bias: torch.Tensor,
+                 diag_offset: Optional[int], quant: str = "bf16",
+                 qcache=None, g_slab: Optional[torch.Tensor] = None,
+                 gt_slab: Optional[torch.Tensor] = None, col0: int = 0,
+                 out3: Optional[torch.Tensor] = None):
+    """Fused forward that also emits the g slab and both scalar partials.
+
+    Returns ``(out3, g_slab, gt_slab)`` where ``out3`` is float32
+    ``{loss, Σ g·dot, Σ g}`` (atomically accumulated — pass the same buffer
+    across chunk calls to sum them), ``g_slab`` is the ``(b, n)`` dL/dlogit
+    slab (bf16, or e4m3 ×448 for fp8/mixed along with its ``(n, b)``
+    transpose ``gt_slab``).  When ``g_slab`` is supplied the chunk is written
+    at column offset ``col0`` with the slab's width as row stride — the ring
+    strategy assembles one ``(b, W·b)`` slab chunk by chunk.
+    """
+    lib = _require_lib()
+    _validate(zimg, ztxt, quant)
+    b, d = zimg.shape
+    n = ztxt.shape[0]
+    dev = zimg.device
+    tp = _prep_scalar(t_prime, dev)
+    bp = _prep_scalar(bias, dev)
+    fp8g = quant in ("fp8", "mixed")
+    if fp8g and b % 4 != 0:
+        raise RuntimeError(f"{quant} fwd+g requires batch % 4 == 0")
+    esz = 1 if fp8g else 2
+    g_dtype = torch.float8_e4m3fn if fp8g else torch.bfloat16
+    if g_slab is None:
+        g_slab = torch.empty((b, n), device=dev, dtype=g_dtype)
+    ldg = g_slab.shape[1]
+    g_ptr = g_slab.data_ptr() + col0 * esz
+    gt_ptr = 0
+    if fp8g:
+        if gt_slab is None:
+            gt_slab = torch.empty((n, b), device=dev, dtype=g_dtype)
+        gt_ptr = gt_slab.data_ptr() + col0 * b * esz
+    if b * ldg * esz >= 2 ** 32:
+        raise RuntimeError(
+            "g slab exceeds the kernel's 32-bit addressing; use the "
+            "recompute path (save_g_enabled would have said no)")
+    if out3 is None:
+        out3 = torch.zeros(3, device=dev, dtype=torch.float32)
+    if quant == "fp8":
+        zi_q, si, zt_q, st = (qcache if qcache is not None
+                              else quantize_fp8_pair(zimg, ztxt))
+        tp_k = tp + si.log() + st.log()
+        fn = lib.siglip_fwdg_fp8
+        zi_ptr, zt_ptr = zi_q.data_ptr(), zt_q.data_ptr()
+    else:
+        tp_k = tp
+        fn = lib.siglip_fwdg_mixed if quant == "mixed" else lib.siglip_fwdg_bf16
+        zi_ptr, zt_ptr = zimg.data_ptr(), ztxt.data_ptr()
+    stream = torch.cuda.current_stream(dev).cuda_stream
+    diag = _DIAG_NONE if diag_offset is None else int(diag_offset)
+    if fp8g:
+        rc = fn(ctypes.c_void_p(stream), ctypes.c_void_p(zi_ptr),
+                ctypes.c_void_p(zt_ptr), ctypes.c_void_p(tp_k.data_ptr()),
+                ctypes.c_void_p(bp.data_ptr()),
+                ctypes.c_void_p(out3.data_ptr()), ctypes.c_void_p(g_ptr),
+                ctypes.c_void_p(gt_ptr), b, n, d, ldg, diag, _kernel_flags())
+    else:
+        rc = fn(ctypes.c_void_p(stream), ctypes.c_void_p(zi_ptr),
+                ctypes.c_void_p(zt_ptr), ctypes.c_void_p(tp_k.data_ptr()),
+                ctypes.c_void_p(bp.data_ptr()),
+                ctypes.c_void_p(out3.data_ptr()), ctypes.c_void_p(g_ptr),
+                b, n, d, ldg, diag, _kernel_flags())
+    _check(rc, "siglip_fwdg")
+    return out3, g_slab, gt_slab if fp8g else None
+
+
+def siglip_bwd_from_g(zimg: torch.Tensor, ztxt: torch.Tensor,
+                      t_prime: torch.Tensor, bias: torch.Tensor,
+                      grad_output: torch.Tensor, out3: torch.Tensor,
+                      g: torch.Tensor, gt: Optional[torch.Tensor],
+                      quant: str = "bf16", qcache=None, on_dztxt=None):
+    """Backward from the fwd+g saved slab: two gradient GEMMs plus the scalar
+    grads — no logits recompute.  Same return contract as :func:`siglip_bwd`.
+
+    ``on_dztxt(dztxt)`` fires as soon as the text gradient exists so a
+    distributed caller overlaps its reduce-scatter with the dzimg GEMM.
+    """
+    b, d = zimg.shape
+    n = ztxt.shape[0]
+    dev = zimg.device
+    tp = _prep_scalar(t_prime, dev)
+    t_true = tp.exp()
+    go = grad_output.detach().reshape(()).to(device=dev, dtype=torch.float32)
+    scale = go * t_true
+    fp8g = quant in ("fp8", "mixed")
+    if fp8g:
+        zi_q, si, zt_q, st = qcache
+        use_mm8 = (b % 16 == 0 and n % 16 == 0 and d % 16 == 0
+                   and hasattr(torch, "_scaled_mm"))
+        if use_mm8:
+            dztxt = scaled_mm8(gt, zi_q, (scale / 448.0) * si)
+            if on_dztxt is not None:
+                on_dztxt(dztxt)
+            dzimg = scaled_mm8(g, zt_q, (scale / 448.0) * st)
+        else:
+            g16 = g.to(torch.bfloat16) * (1.0 / 448.0)
+            dztxt = (g16.T @ zimg) * scale
+            if on_dztxt is not None:
+                on_dztxt(dztxt)
+            dzimg = (g16 @ ztxt) * scale
+        t_eff = t_true * si * st if quant == "fp8" else t_true
+    else:
+        dztxt = (g.T @ zimg) * scale
+        if on_dztxt is not None:
+            on_dztxt(dztxt)
+        dzimg = (g @ ztxt) * scale
+        t_eff = t_true
+    dt_prime = (out3[1] * go * t_eff).to(t_prime.dtype).reshape(t_prime.shape)
+    dbias = (out3[2] * go).to(bias.dtype).reshape(bias.shape)
+    return (dzimg.to(zimg.dtype), dztxt.to(ztxt.dtype), dt_prime, dbias)

@@ -17,7 +17,15 @@
 // the two rocBLAS GEMMs (dzimg = t·g@ztxt, dztxt = t·gᵀ@zimg) on the Python
 // side (ops/__init__.py).
 //
-// Kernel structure (templates: MODE 0 = fwd, 1 = bwd-g; EB = element bytes,
+// MODE 2 ("fwd+g") fuses the two: one kernel emits the loss, the g slab and
+// both scalar partials, so a training step never computes the logits GEMM
+// twice — backward is then pure GEMMs on the saved slab (the reference's
+// autograd does exactly this by saving the logits graph; here the saved
+// state is the (b,n) g slab, e.g. 2 GiB at B=32k bf16, trivial against
+// 288 GB HBM3E).  The recompute MODE 1 remains for the huge-batch chunked
+// path where the slab would not fit (BASELINE config 3).
+//
+// Kernel structure (templates: MODE 0 = fwd, 1 = bwd-g, 2 = fwd+g; EB = element bytes,
 // 2 = bf16 via v_mfma_f32_16x16x32_bf16, 1 = fp8-e4m3 via the MX-scaled
 // v_mfma_scale_f32_16x16x128_f8f6f4 with unit block scales — 2× the bf16
 // MFMA rate and half the staged bytes; per-tensor scales are folded into the
@@ -189,7 +197,8 @@ __device__ __forceinline__ void tile_body(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     float t, float bias, float* __restrict__ out, __bf16* __restrict__ g_out,
     unsigned char* __restrict__ gt_out,
-    int b, int n, int d, int diag, int row_base, int col_base, char* smem) {
+    int b, int n, int d, int ldg, int diag, int row_base, int col_base,
+    char* smem) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int wrow = (wave >> 2) * 128;   // wave sub-tile origin: 2×4 grid
@@ -331,7 +340,13 @@ __device__ __forceinline__ void tile_body(
 
   // Epilogue.  C/D layout (shape-determined, dtype-independent on gfx950):
   //   col = lane&15, row = (lane>>4)*4 + reg.
-  float s0 = 0.f, s1 = 0.f;
+  // Per element: x = −l·z.  loss term = softplus(x); g = −l·σ(−l·z) =
+  // ∓σ(x).  softplus and sigmoid share one exp: with y = e^{−|x|},
+  // softplus(x) = max(x,0) + log(1+y) and σ(x) = (x≥0 ? 1 : y)/(1+y).
+  // MODE 0 emits only the loss; MODE 1 only g + scalar partials; MODE 2
+  // ("fwd+g") all three — the whole fwd+bwd recompute collapse costs one
+  // extra log and the loss adds per lane.
+  float s_loss = 0.f, s_gdot = 0.f, s_g = 0.f;
   if (MODE == 0) {
 #pragma unroll
     for (int mi = 0; mi < FM; ++mi) {
@@ -344,16 +359,18 @@ __device__ __forceinline__ void tile_body(
           if (INTERIOR || (grow < b && gcol < n)) {
             const float z = acc[mi][ni][reg] * t + bias;
             const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
-            s0 += softplus_f(pos ? -z : z);
+            s_loss += softplus_f(pos ? -z : z);
           }
         }
       }
     }
   } else {
     // One per-lane base offset + per-(mi,reg) scalar row offset keeps the
-    // store addressing affine — per-element (size_t)grow*n math made the
+    // store addressing affine — per-element (size_t)grow*ldg math made the
     // allocator hoist 128 addresses and spill.  Caller guarantees
-    // b*n*esz < 2^32 (ops/__init__.py column-chunks the slab).
+    // b*ldg*esz < 2^32 (ops/__init__.py column-chunks the slab).  ldg ≥ n
+    // is the slab row stride — the ring strategy writes each received
+    // chunk's g at its column offset inside one (b, W·b) slab.
     // bf16 path writes one bf16 slab; fp8 path writes e4m3 at a FIXED ×448
     // scale (|g| ≤ 1 by construction) AND a transposed (n, b) slab — both
     // consumed by torch._scaled_mm (mat1 must be row-major; a Python-side
@@ -361,9 +378,9 @@ __device__ __forceinline__ void tile_body(
     // 4 accumulator regs of a fragment are 4 consecutive gᵀ columns → one
     // packed 4-byte store.
     char* gb = reinterpret_cast<char*>(g_out) +
-        ((size_t)row_base * n + col_base) * EB_G;
+        ((size_t)row_base * ldg + col_base) * EB_G;
     const unsigned lane_off =
-        (unsigned)(wrow + (lane >> 4) * 4) * (unsigned)n
+        (unsigned)(wrow + (lane >> 4) * 4) * (unsigned)ldg
         + (unsigned)(wcol + (lane & 15));
     unsigned char* gtb = gt_out +
         ((size_t)col_base * b + row_base);
@@ -383,9 +400,14 @@ __device__ __forceinline__ void tile_body(
             const float dot = acc[mi][ni][reg];
             const float z = dot * t + bias;
             const bool pos = (diag != DIAG_NONE) && (gcol == grow + diag);
-            const float gv = sigmoid_fast(pos ? z : -z);
-            const float g = pos ? -gv : gv;
-            const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)n;
+            const float x = pos ? -z : z;            // −l·z
+            const float y = __expf(-fabsf(x));
+            const float r = __builtin_amdgcn_rcpf(1.0f + y);
+            if (MODE == 2)
+              s_loss += fmaxf(x, 0.0f) + __logf(1.0f + y);
+            const float sig = (x >= 0.0f) ? r : y * r;   // σ(x)
+            const float g = pos ? -sig : sig;
+            const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)ldg;
             if (EB_G == 1) {
               const unsigned char q = __hip_fp8_e4m3(g * 448.0f).__x;
               reinterpret_cast<unsigned char*>(gb)[
@@ -395,8 +417,8 @@ __device__ __forceinline__ void tile_body(
               reinterpret_cast<__bf16*>(gb)[lane_off + row_off + ni * 16] =
                   (__bf16)g;
             }
-            s0 += g * dot;
-            s1 += g;
+            s_gdot += g * dot;
+            s_g += g;
           }
         }
         if (EB_G == 1) {
@@ -421,12 +443,25 @@ __device__ __forceinline__ void tile_body(
 
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) {
-    s0 += __shfl_down(s0, off);
-    if (MODE == 1) s1 += __shfl_down(s1, off);
+    if (MODE != 1) s_loss += __shfl_down(s_loss, off);
+    if (MODE != 0) {
+      s_gdot += __shfl_down(s_gdot, off);
+      s_g += __shfl_down(s_g, off);
+    }
   }
   if (lane == 0) {
-    atomicAdd(&out[0], s0);
-    if (MODE == 1) atomicAdd(&out[1], s1);
+    // out layout: MODE 0 → {loss}; MODE 1 → {Σg·dot, Σg};
+    //             MODE 2 → {loss, Σg·dot, Σg}.
+    if (MODE == 0) {
+      atomicAdd(&out[0], s_loss);
+    } else if (MODE == 1) {
+      atomicAdd(&out[0], s_gdot);
+      atomicAdd(&out[1], s_g);
+    } else {
+      atomicAdd(&out[0], s_loss);
+      atomicAdd(&out[1], s_gdot);
+      atomicAdd(&out[2], s_g);
+    }
   }
 }
 
@@ -465,14 +500,14 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
     float* __restrict__ out, __bf16* __restrict__ g_out,
     unsigned char* __restrict__ gt_out,
-    int b, int n, int d, int diag, int flags) {
+    int b, int n, int d, int ldg, int diag, int flags) {
   __shared__ char smem[4 * TILE_BYTES];
   int bx, by;
   remap_block(flags, bx, by);
   const float t = __expf(*t_prime);
   const float bias = *bias_p;
   tile_body<MODE, true, EB, EB_G>(zimg, ztxt, t, bias, out, g_out, gt_out,
-                                  b, n, d, diag, bx * BM, by * BN, smem);
+                                  b, n, d, ldg, diag, bx * BM, by * BN, smem);
 }
 
 // General kernel: interior blocks take the DMA fast path, edge blocks the
@@ -484,7 +519,7 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
     float* __restrict__ out, __bf16* __restrict__ g_out,
     unsigned char* __restrict__ gt_out,
-    int b, int n, int d, int diag, int flags) {
+    int b, int n, int d, int ldg, int diag, int flags) {
   __shared__ char smem[4 * TILE_BYTES];
   int bx, by;
   remap_block(flags, bx, by);
@@ -498,10 +533,11 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
       (d % (128 / EB) == 0) && (n % 8 == 0);
   if (interior)
     tile_body<MODE, true, EB, EB_G>(zimg, ztxt, t, bias, out, g_out, gt_out,
-                                    b, n, d, diag, row_base, col_base, smem);
+                                    b, n, d, ldg, diag, row_base, col_base,
+                                    smem);
   else
     tile_body<MODE, false, EB, EB_G>(zimg, ztxt, t, bias, out, g_out,
-                                     gt_out, b, n, d, diag, row_base,
+                                     gt_out, b, n, d, ldg, diag, row_base,
                                      col_base, smem);
 }
 
@@ -510,9 +546,9 @@ inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 template <int MODE, int EB, int EB_G = EB>
 int launch(uintptr_t stream, const void* zimg, const void* ztxt,
            const void* t_prime, const void* bias, void* out, void* g_out,
-           void* gt_out, int b, int n, int d, int diag, int flags) {
+           void* gt_out, int b, int n, int d, int ldg, int diag, int flags) {
   if (b <= 0 || n <= 0) return (int)hipSuccess;
-  if (d % (16 / EB) != 0) return (int)hipErrorInvalidValue;
+  if (d % (16 / EB) != 0 || ldg < n) return (int)hipErrorInvalidValue;
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
   const bool interior = (b % BM == 0) && (n % BN == 0) &&
       (d % (128 / EB) == 0);
@@ -522,7 +558,7 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
                        (const char*)zimg, (const char*)ztxt,
                        (const float*)t_prime, (const float*)bias,
                        (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
-                       b, n, d, diag, flags);
+                       b, n, d, ldg, diag, flags);
   else
     hipLaunchKernelGGL((siglip_tile_kernel<MODE, EB, EB_G>), grid,
                        dim3(THREADS),
@@ -530,7 +566,7 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
                        (const char*)zimg, (const char*)ztxt,
                        (const float*)t_prime, (const float*)bias,
                        (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
-                       b, n, d, diag, flags);
+                       b, n, d, ldg, diag, flags);
   return (int)hipGetLastError();
 }
 
@@ -538,27 +574,27 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
 
 extern "C" {
 
-int siglip_ext_abi(void) { return 5; }
+int siglip_ext_abi(void) { return 6; }
 
 int siglip_fwd_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                     const void* t_prime, const void* bias, void* loss_out,
                     int b, int n, int d, int diag, int flags) {
   return launch<0, 2>(stream, zimg, ztxt, t_prime, bias, loss_out, nullptr,
-                      nullptr, b, n, d, diag, flags);
+                      nullptr, b, n, d, n, diag, flags);
 }
 
 int siglip_bwd_g_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                       const void* t_prime, const void* bias, void* g_out,
                       void* scal, int b, int n, int d, int diag, int flags) {
   return launch<1, 2>(stream, zimg, ztxt, t_prime, bias, scal, g_out,
-                      nullptr, b, n, d, diag, flags);
+                      nullptr, b, n, d, n, diag, flags);
 }
 
 int siglip_fwd_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
                    const void* t_prime, const void* bias, void* loss_out,
                    int b, int n, int d, int diag, int flags) {
   return launch<0, 1>(stream, zimg, ztxt, t_prime, bias, loss_out, nullptr,
-                      nullptr, b, n, d, diag, flags);
+                      nullptr, b, n, d, n, diag, flags);
 }
 
 // Mixed policy: bf16 logits recompute, e4m3 ×448 g and gᵀ slabs (for the
@@ -569,7 +605,7 @@ int siglip_bwd_g_mixed(uintptr_t stream, const void* zimg, const void* ztxt,
                        int diag, int flags) {
   if (b % 4 != 0) return (int)hipErrorInvalidValue;
   return launch<1, 2, 1>(stream, zimg, ztxt, t_prime, bias, scal, g_out,
-                         gt_out, b, n, d, diag, flags);
+                         gt_out, b, n, d, n, diag, flags);
 }
 
 // fp8 backward emits g (b,n) AND its transpose gt (n,b), both e4m3 ×448.
@@ -579,7 +615,41 @@ int siglip_bwd_g_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
                      int flags) {
   if (b % 4 != 0) return (int)hipErrorInvalidValue;  // packed 4-B gt stores
   return launch<1, 1>(stream, zimg, ztxt, t_prime, bias, scal, g_out,
-                      gt_out, b, n, d, diag, flags);
+                      gt_out, b, n, d, n, diag, flags);
+}
+
+// ---- fwd+g ("saved-g") entry points: one kernel computes the loss, the g
+// slab and both scalar partials — backward then runs only the two gradient
+// GEMMs on the saved slab (no logits recompute).  `out` is float[3] =
+// {loss, Σg·dot, Σg}; `ldg` is the g-slab row stride in elements (≥ n) so
+// the ring strategy can write each chunk at its column offset inside one
+// (b, W·b) slab.
+
+int siglip_fwdg_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
+                     const void* t_prime, const void* bias, void* out,
+                     void* g_out, int b, int n, int d, int ldg, int diag,
+                     int flags) {
+  return launch<2, 2>(stream, zimg, ztxt, t_prime, bias, out, g_out,
+                      nullptr, b, n, d, ldg, diag, flags);
+}
+
+// mixed: bf16 logits, e4m3 ×448 g and gᵀ slabs for the fp8 grad GEMMs.
+int siglip_fwdg_mixed(uintptr_t stream, const void* zimg, const void* ztxt,
+                      const void* t_prime, const void* bias, void* out,
+                      void* g_out, void* gt_out, int b, int n, int d,
+                      int ldg, int diag, int flags) {
+  if (b % 4 != 0) return (int)hipErrorInvalidValue;
+  return launch<2, 2, 1>(stream, zimg, ztxt, t_prime, bias, out, g_out,
+                         gt_out, b, n, d, ldg, diag, flags);
+}
+
+int siglip_fwdg_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
+                    const void* t_prime, const void* bias, void* out,
+                    void* g_out, void* gt_out, int b, int n, int d,
+                    int ldg, int diag, int flags) {
+  if (b % 4 != 0) return (int)hipErrorInvalidValue;
+  return launch<2, 1>(stream, zimg, ztxt, t_prime, bias, out, g_out,
+                      gt_out, b, n, d, ldg, diag, flags);
 }
 
 }  // extern "C"
