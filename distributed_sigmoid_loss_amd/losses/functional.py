@@ -177,9 +177,10 @@ class _FusedSigmoidLoss(torch.autograd.Function):
         save_g = (want_grad and col_chunk is None and zimg.is_cuda
                   and ops.save_g_enabled(b, n, quant))
         if save_g:
-            out3, g, gt = ops.siglip_fwd_g(zimg, ztxt, t_prime, bias,
-                                           diag_offset, quant=quant,
-                                           qcache=qc)
+            buf, g, gt = ops.siglip_fwd_g(zimg, ztxt, t_prime, bias,
+                                          diag_offset, quant=quant,
+                                          qcache=qc)
+            out3 = ops.reduce_out3(buf)
             loss = out3[0].clone()
             saved = (zimg, ztxt, t_prime, bias, out3, g) \
                 + ((gt,) if gt is not None else ()) \

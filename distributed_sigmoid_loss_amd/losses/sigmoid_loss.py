@@ -121,7 +121,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
             if quant != "fp8":
                 g_slab = torch.empty((b_img, world * b_txt),
                                      device=zimg.device, dtype=g_dtype)
-                out3 = torch.zeros(3, device=zimg.device, dtype=torch.float32)
+                out3 = _ops._out_buf(zimg.device)
             if quant in ("fp8", "mixed"):
                 gt_slab = torch.empty((world * b_txt, b_img),
                                       device=zimg.device, dtype=g_dtype)
@@ -139,11 +139,11 @@ class _RingAllGatherLoss(torch.autograd.Function):
                     qcache=qc, gt_slab=gt_slab, col0=src * b_txt)
                 g_chunks[src] = g_c
                 out3s[src] = o3
-                return o3[0]
-            o3, _, _ = _ops.siglip_fwd_g(
+                return None
+            _ops.siglip_fwd_g(
                 zimg, zt_chunk, t_prime, bias, diag, quant=quant,
                 g_slab=g_slab, gt_slab=gt_slab, col0=src * b_txt, out3=out3)
-            return o3[0]
+            return None   # shared out3 accumulates across chunks
 
         if world > 1:
             left = (rank - 1 + world) % world
@@ -167,10 +167,19 @@ class _RingAllGatherLoss(torch.autograd.Function):
                     handle = neighbour_exchange_start(left, right, recv,
                                                       group=group)
                 with roctx_range(f"ring_chunk{hop}_loss"):
-                    loss = loss + chunk_fwd(recv, src, None)
+                    part = chunk_fwd(recv, src, None)
+                    if part is not None:
+                        loss = loss + part
 
         if save_g:
-            loss = loss.clone()   # detach from the shared out3 views
+            # Reduce the per-XCD scalar buffers once, after every chunk's
+            # kernel has accumulated into them.
+            if quant == "fp8":
+                out3s = [_ops.reduce_out3(o) for o in out3s]
+                loss = sum(o[0] for o in out3s).clone()
+            else:
+                out3 = _ops.reduce_out3(out3)
+                loss = out3[0].clone()
 
         saved = [zimg, t_prime, bias] + chunks
         idx = {}

@@ -32,6 +32,23 @@ from .build import SO_PATH, build as build_extension
 
 _DIAG_NONE = -(2 ** 31)
 
+# Kernel scalar outputs land in a zeroed (8, 32)-float buffer — one 128-B
+# cache-line slot per XCD (per-block atomics stay XCD-local; a single shared
+# line measured as a serialized ~10ns/op drain).  Slot layout:
+# [0] loss, [1] Σg·dot, [2] Σg.  Reduce with the helpers below.
+_OUT_SLOTS, _OUT_STRIDE = 8, 32
+
+
+def _out_buf(device) -> torch.Tensor:
+    return torch.zeros((_OUT_SLOTS, _OUT_STRIDE), device=device,
+                       dtype=torch.float32)
+
+
+def reduce_out3(buf: torch.Tensor) -> torch.Tensor:
+    """Sum the per-XCD slots of a kernel scalar-output buffer →
+    ``[loss, Σg·dot, Σg]`` (whichever the mode wrote; others are 0)."""
+    return buf[:, :3].sum(dim=0)
+
 
 def _kernel_flags() -> int:
     # bit 0: XCD-contiguous block remap — default ON: together with the
@@ -85,9 +102,9 @@ def _load():
         [ctypes.c_void_p] * 8 + [ctypes.c_int] * 6)
     lib.siglip_fwdg_fp8.restype = ctypes.c_int
     lib.siglip_fwdg_fp8.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 6
-    if lib.siglip_ext_abi() != 6:
+    if lib.siglip_ext_abi() != 7:
         _lib_err = (f"stale HIP extension at {SO_PATH} "
-                    f"(ABI {lib.siglip_ext_abi()}, need 6); rebuild with: "
+                    f"(ABI {lib.siglip_ext_abi()}, need 7); rebuild with: "
                     "python -m distributed_sigmoid_loss_amd.ops.build --force")
         return None
     _lib = lib
@@ -174,16 +191,16 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     else:
         zi_ptr, zt_ptr = zimg.data_ptr(), ztxt.data_ptr()
         fn = lib.siglip_fwd_bf16
-    loss = torch.zeros((), device=dev, dtype=torch.float32)
+    buf = _out_buf(dev)
     stream = torch.cuda.current_stream(dev).cuda_stream
     diag = _DIAG_NONE if diag_offset is None else int(diag_offset)
     _check(fn(
         ctypes.c_void_p(stream),
         ctypes.c_void_p(zi_ptr), ctypes.c_void_p(zt_ptr),
         ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
-        ctypes.c_void_p(loss.data_ptr()), b, n, d, diag, _kernel_flags()),
+        ctypes.c_void_p(buf.data_ptr()), b, n, d, diag, _kernel_flags()),
         "siglip_fwd")
-    return loss
+    return buf[:, 0].sum()
 
 
 def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
@@ -229,7 +246,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         tp_k = tp
         g_fn = lib.siglip_bwd_g_bf16
         zi_k, zt_k = zimg, ztxt
-    scal = torch.zeros(2, device=dev, dtype=torch.float32)
+    scal = _out_buf(dev)
     stream = torch.cuda.current_stream(dev).cuda_stream
 
     # Column slab sizing: the kernel's g-store addressing is 32-bit, so
@@ -357,8 +374,9 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     dztxt = dztxt.to(ztxt.dtype)
     # dt' = go · Σ g·(z−bias) = go · t_eff · Σ g·dot_q  (t_eff ≡ t for bf16).
     t_eff = tp_k.exp()
-    dt_prime = (scal[0] * go * t_eff).to(t_prime.dtype).reshape(t_prime.shape)
-    dbias = (scal[1] * go).to(bias.dtype).reshape(bias.shape)
+    sv = reduce_out3(scal)
+    dt_prime = (sv[1] * go * t_eff).to(t_prime.dtype).reshape(t_prime.shape)
+    dbias = (sv[2] * go).to(bias.dtype).reshape(bias.shape)
     return dzimg, dztxt, dt_prime, dbias
 
 
@@ -410,9 +428,10 @@ def siglip_fwd_g(zimg: torch.Tensor, ztxt: torch.Tensor,
                  out3: Optional[torch.Tensor] = None):
     """Fused forward that also emits the g slab and both scalar partials.
 
-    Returns ``(out3, g_slab, gt_slab)`` where ``out3`` is float32
-    ``{loss, Σ g·dot, Σ g}`` (atomically accumulated — pass the same buffer
-    across chunk calls to sum them), ``g_slab`` is the ``(b, n)`` dL/dlogit
+    Returns ``(out3, g_slab, gt_slab)`` where ``out3`` is the RAW (8, 32)
+    per-XCD scalar buffer (atomically accumulated — pass the same buffer
+    across chunk calls to sum them; reduce with :func:`reduce_out3` →
+    ``[loss, Σ g·dot, Σ g]``), ``g_slab`` is the ``(b, n)`` dL/dlogit
     slab (bf16, or e4m3 ×448 for fp8/mixed along with its ``(n, b)``
     transpose ``gt_slab``).  When ``g_slab`` is supplied the chunk is written
     at column offset ``col0`` with the slab's width as row stride — the ring
@@ -444,7 +463,7 @@ def siglip_fwd_g(zimg: torch.Tensor, ztxt: torch.Tensor,
             "g slab exceeds the kernel's 32-bit addressing; use the "
             "recompute path (save_g_enabled would have said no)")
     if out3 is None:
-        out3 = torch.zeros(3, device=dev, dtype=torch.float32)
+        out3 = _out_buf(dev)
     if quant == "fp8":
         zi_q, si, zt_q, st = (qcache if qcache is not None
                               else quantize_fp8_pair(zimg, ztxt))
@@ -479,7 +498,8 @@ def siglip_bwd_from_g(zimg: torch.Tensor, ztxt: torch.Tensor,
                       g: torch.Tensor, gt: Optional[torch.Tensor],
                       quant: str = "bf16", qcache=None, on_dztxt=None):
     """Backward from the fwd+g saved slab: two gradient GEMMs plus the scalar
-    grads — no logits recompute.  Same return contract as :func:`siglip_bwd`.
+    grads — no logits recompute.  ``out3`` is the REDUCED 3-vector
+    (:func:`reduce_out3`).  Same return contract as :func:`siglip_bwd`.
 
     ``on_dztxt(dztxt)`` fires as soon as the text gradient exists so a
     distributed caller overlaps its reduce-scatter with the dzimg GEMM.

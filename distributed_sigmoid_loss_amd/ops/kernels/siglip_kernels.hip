@@ -86,6 +86,8 @@ constexpr int HALF_BYTES = BM * HROW;        // 16 KiB per operand K-half
 constexpr int TILE_BYTES = 2 * HALF_BYTES;   // 32 KiB per operand K-step
 constexpr int DIAG_NONE = INT_MIN;
 constexpr int NXCD = 8;
+constexpr int OUT_STRIDE = 32;   // floats per XCD slot of the scalar-output
+                                 // buffer (128 B = one cache line)
 constexpr int GROUP_M = 8;       // block-rows per locality group (bit1)
 
 // softplus via the inlined fast log/exp (log1pf is a device-lib CALL —
@@ -449,18 +451,40 @@ __device__ __forceinline__ void tile_body(
       s_g += __shfl_down(s_g, off);
     }
   }
+  // Cross-wave reduction in LDS (the operand tiles are dead — every wave's
+  // MFMA reads finished before its epilogue), then ONE atomic per scalar
+  // per BLOCK into the block's XCD slot of the [8][32]-float output buffer.
+  // Per-WAVE atomics to a single global line measured as a serialized
+  // ~10 ns/op drain that dominated the fwd+g wall (3 × 131k ops ≈ +3 ms at
+  // B=32k — all three fwd+g dtypes converged to the same ~5.7 ms floor);
+  // per-block per-XCD 128-B slots cut the per-line rate ~384× and the
+  // Python side sums the 256 floats once.
+  __syncthreads();
+  float* red = reinterpret_cast<float*>(smem);
   if (lane == 0) {
-    // out layout: MODE 0 → {loss}; MODE 1 → {Σg·dot, Σg};
-    //             MODE 2 → {loss, Σg·dot, Σg}.
-    if (MODE == 0) {
-      atomicAdd(&out[0], s_loss);
-    } else if (MODE == 1) {
-      atomicAdd(&out[0], s_gdot);
-      atomicAdd(&out[1], s_g);
-    } else {
-      atomicAdd(&out[0], s_loss);
-      atomicAdd(&out[1], s_gdot);
-      atomicAdd(&out[2], s_g);
+    red[wave] = s_loss;
+    red[8 + wave] = s_gdot;
+    red[16 + wave] = s_g;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float r_loss = 0.f, r_gdot = 0.f, r_g = 0.f;
+#pragma unroll
+    for (int w = 0; w < 8; ++w) {
+      r_loss += red[w];
+      r_gdot += red[8 + w];
+      r_g += red[16 + w];
+    }
+    // Slot by the PRE-remap flat id: hardware dispatches block b to XCD
+    // b % 8, so slots are XCD-local lines.
+    const int slot =
+        ((blockIdx.y * gridDim.x + blockIdx.x) & (NXCD - 1)) * OUT_STRIDE;
+    // out layout per slot: [0] loss (MODE 0/2), [1] Σg·dot, [2] Σg
+    // (MODE 1/2).
+    if (MODE != 1) atomicAdd(&out[slot + 0], r_loss);
+    if (MODE != 0) {
+      atomicAdd(&out[slot + 1], r_gdot);
+      atomicAdd(&out[slot + 2], r_g);
     }
   }
 }
@@ -574,7 +598,7 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
 
 extern "C" {
 
-int siglip_ext_abi(void) { return 6; }
+int siglip_ext_abi(void) { return 7; }
 
 int siglip_fwd_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                     const void* t_prime, const void* bias, void* loss_out,
@@ -620,8 +644,9 @@ int siglip_bwd_g_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
 
 // ---- fwd+g ("saved-g") entry points: one kernel computes the loss, the g
 // slab and both scalar partials — backward then runs only the two gradient
-// GEMMs on the saved slab (no logits recompute).  `out` is float[3] =
-// {loss, Σg·dot, Σg}; `ldg` is the g-slab row stride in elements (≥ n) so
+// GEMMs on the saved slab (no logits recompute).  `out` is a zeroed
+// float[8][32] per-XCD-slot buffer; slot layout {loss, Σg·dot, Σg} — the
+// caller sums slots.  `ldg` is the g-slab row stride in elements (≥ n) so
 // the ring strategy can write each chunk at its column offset inside one
 // (b, W·b) slab.
 

@@ -58,7 +58,8 @@ def ref_g_and_scalars(zi, zt, tp, bs, diag):
 def test_fwdg_matches_reference(b, n, diag):
     d = 768 if b % 8 == 0 else 256
     zi, zt, tp, bs = make_inputs(b, n, d, seed=b + n)
-    out3, g, gt = ops.siglip_fwd_g(zi, zt, tp, bs, diag)
+    buf, g, gt = ops.siglip_fwd_g(zi, zt, tp, bs, diag)
+    out3 = ops.reduce_out3(buf)
     r_loss, r_g, r_gdot, r_gsum = ref_g_and_scalars(zi, zt, tp, bs, diag)
     torch.cuda.synchronize()
     assert gt is None
@@ -75,16 +76,17 @@ def test_fwdg_slab_offset_write():
     b, n, d = 256, 256, 768
     zi, zt, tp, bs = make_inputs(b, 2 * n, d, seed=3)
     # one call over the full width
-    out3_full, g_full, _ = ops.siglip_fwd_g(zi, zt, tp, bs, 64)
-    # two chunked calls into a shared slab + shared out3
+    buf_full, g_full, _ = ops.siglip_fwd_g(zi, zt, tp, bs, 64)
+    # two chunked calls into a shared slab + shared scalar buffer
     g_slab = torch.empty((b, 2 * n), device="cuda", dtype=torch.bfloat16)
-    out3 = torch.zeros(3, device="cuda", dtype=torch.float32)
+    buf = ops._out_buf("cuda")
     ops.siglip_fwd_g(zi, zt[:n], tp, bs, 64, g_slab=g_slab, col0=0,
-                     out3=out3)
+                     out3=buf)
     ops.siglip_fwd_g(zi, zt[n:], tp, bs, 64 - n, g_slab=g_slab, col0=n,
-                     out3=out3)
+                     out3=buf)
     torch.cuda.synchronize()
-    assert torch.allclose(out3, out3_full, rtol=1e-3, atol=1e-3)
+    assert torch.allclose(ops.reduce_out3(buf), ops.reduce_out3(buf_full),
+                          rtol=1e-3, atol=1e-3)
     assert torch.equal(g_slab, g_full)
 
 
@@ -160,7 +162,10 @@ def test_fwdg_quant_vs_fp32(quant):
     ref.backward()
     torch.cuda.synchronize()
 
-    rtol, atol = (1e-1, 2e-2) if quant == "fp8" else (5e-2, 5e-3)
+    # Embedding grads pass through the e4m3 grad GEMMs in BOTH policies
+    # (mixed differs only in the logits precision), so both get fp8-class
+    # tolerances on grads; mixed's loss is bf16-exact.
+    rtol, atol = 1e-1, 2e-2
     assert torch.allclose(loss.float(), ref, rtol=5e-2)
     assert torch.allclose(zi.grad.float(), zi2.grad, rtol=rtol, atol=atol)
     assert torch.allclose(zt.grad.float(), zt2.grad, rtol=rtol, atol=atol)

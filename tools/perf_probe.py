@@ -79,7 +79,7 @@ def main():
     lib = ops._require_lib()
     c = min(args.col_chunk, b)
     g = torch.empty((b, c), device=dev, dtype=torch.bfloat16)
-    scal = torch.zeros(2, device=dev, dtype=torch.float32)
+    scal = ops._out_buf(dev)
     ztc = zt[:c].contiguous()
     stream = torch.cuda.current_stream().cuda_stream
 
@@ -106,6 +106,32 @@ def main():
     t_bwd = time_fn(lambda: ops.siglip_bwd(zi, zt, tp, bs, 0, go,
                                            args.col_chunk))
     print(f"bwd total (wrapper)   : {t_bwd:8.3f} ms")
+
+    # fwd+g (saved-g) pieces: the fused fwd emits loss + g slab + scalars;
+    # backward is two full-slab GEMMs.
+    t_fwdg = time_fn(lambda: ops.siglip_fwd_g(zi, zt, tp, bs, 0))
+    print(f"fwd+g fused kernel    : {t_fwdg:8.3f} ms   "
+          f"{flops_fwd / t_fwdg / 1e9:7.1f} TF/s")
+    buf3, gfull, _ = ops.siglip_fwd_g(zi, zt, tp, bs, 0)
+    out3 = ops.reduce_out3(buf3)
+    t_bfg = time_fn(lambda: ops.siglip_bwd_from_g(zi, zt, tp, bs, go, out3,
+                                                  gfull, None))
+    print(f"bwd_from_g (2 GEMMs)  : {t_bfg:8.3f} ms")
+    t_mmA = time_fn(lambda: gfull @ zt)
+    t_mmB = time_fn(lambda: gfull.T @ zi)
+    print(f"GEMM g@zt  full slab  : {t_mmA:8.3f} ms   "
+          f"{2.0 * b * b * d / t_mmA / 1e9:7.1f} TF/s")
+    print(f"GEMM gT@zi full slab  : {t_mmB:8.3f} ms   "
+          f"{2.0 * b * b * d / t_mmB / 1e9:7.1f} TF/s")
+    del buf3, out3, gfull
+    t_fwdg8 = time_fn(lambda: ops.siglip_fwd_g(zi, zt, tp, bs, 0,
+                                               quant="fp8"))
+    print(f"fwd+g fp8 (incl quant): {t_fwdg8:8.3f} ms   "
+          f"{flops_fwd / t_fwdg8 / 1e9:7.1f} TF/s")
+    t_fwdgm = time_fn(lambda: ops.siglip_fwd_g(zi, zt, tp, bs, 0,
+                                               quant="mixed"))
+    print(f"fwd+g mixed           : {t_fwdgm:8.3f} ms   "
+          f"{flops_fwd / t_fwdgm / 1e9:7.1f} TF/s")
 
     # Whole training step as the bench runs it.
     model = TwoTowerModel(d, d).to(device=dev, dtype=torch.bfloat16)
