@@ -56,11 +56,15 @@ def _kernel_flags() -> int:
     #        and profiles/README.md); SIGLIP_XCD_SWZ=0 disables.
     # bit 1: grouped block walk for L2 panel reuse (default on;
     #        SIGLIP_GROUP_SWZ=0 disables for A/B profiling).
+    # bit 2: non-temporal g/gᵀ slab stores (SIGLIP_NT_G=1; default off —
+    #        measure before enabling).
     f = 0
     if os.environ.get("SIGLIP_XCD_SWZ", "1") != "0":
         f |= 1
     if os.environ.get("SIGLIP_GROUP_SWZ", "1") != "0":
         f |= 2
+    if os.environ.get("SIGLIP_NT_G", "0") == "1":
+        f |= 4
     gm = os.environ.get("SIGLIP_GROUP_M", "4")
     f |= {"8": 0, "1": 1, "4": 2, "16": 3}.get(gm, 0) << 4
     return f

@@ -137,6 +137,14 @@ __device__ __forceinline__ void stage_guarded(const char* __restrict__ gsrc,
   }
 }
 
+// Optionally non-temporal store (flag bit 2): the g/gᵀ slabs are consumed
+// later by the gradient GEMMs from L3/HBM, so nt stores skip the L2 and
+// leave it to the operand panels.
+template <bool NT, typename T>
+__device__ __forceinline__ void st_g(T* p, T v) {
+  if (NT) __builtin_nontemporal_store(v, p); else *p = v;
+}
+
 __device__ __forceinline__ i32x8 pack8(const uint4 lo, const uint4 hi) {
   return i32x8{(int)lo.x, (int)lo.y, (int)lo.z, (int)lo.w,
                (int)hi.x, (int)hi.y, (int)hi.z, (int)hi.w};
@@ -194,7 +202,7 @@ __device__ __forceinline__ void mma_ktile_fp8(const char* smem,
 
 // g-slab element bytes follow the compute dtype: bf16 kernels emit bf16 g,
 // fp8 kernels emit e4m3 g (×448 fixed scale).
-template <int MODE, bool INTERIOR, int EB, int EB_G = EB>
+template <int MODE, bool INTERIOR, int EB, int EB_G = EB, bool NTG = false>
 __device__ __forceinline__ void tile_body(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     float t, float bias, float* __restrict__ out, __bf16* __restrict__ g_out,
@@ -412,12 +420,12 @@ __device__ __forceinline__ void tile_body(
             const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)ldg;
             if (EB_G == 1) {
               const unsigned char q = __hip_fp8_e4m3(g * 448.0f).__x;
-              reinterpret_cast<unsigned char*>(gb)[
-                  lane_off + row_off + ni * 16] = q;
+              st_g<NTG>(reinterpret_cast<unsigned char*>(gb)
+                            + lane_off + row_off + ni * 16, q);
               packed |= (unsigned)q << (8 * reg);
             } else {
-              reinterpret_cast<__bf16*>(gb)[lane_off + row_off + ni * 16] =
-                  (__bf16)g;
+              st_g<NTG>(reinterpret_cast<__bf16*>(gb)
+                            + lane_off + row_off + ni * 16, (__bf16)g);
             }
             s_gdot += g * dot;
             s_g += g;
@@ -427,7 +435,7 @@ __device__ __forceinline__ void tile_body(
           const unsigned toff = lane_off_t + (unsigned)(ni * 16) * (unsigned)b
               + (unsigned)(mi * 16);
           if (INTERIOR) {
-            *reinterpret_cast<unsigned*>(gtb + toff) = packed;
+            st_g<NTG>(reinterpret_cast<unsigned*>(gtb + toff), packed);
           } else {
             const int grow0 = row_base + wrow + mi * 16 + (lane >> 4) * 4;
             const int gcol = col_base + wcol + ni * 16 + (lane & 15);
@@ -518,7 +526,7 @@ __device__ __forceinline__ void remap_block(int flags, int& bx, int& by) {
 // Interior-only kernel: every tile full, d % K-step == 0, n%8==0 — checked
 // by the host launcher.  Separate from the general kernel so the hot path's
 // register allocation is not inflated by the guarded path.
-template <int MODE, int EB, int EB_G = EB>
+template <int MODE, int EB, int EB_G = EB, bool NTG = false>
 __launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
@@ -530,14 +538,15 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
   remap_block(flags, bx, by);
   const float t = __expf(*t_prime);
   const float bias = *bias_p;
-  tile_body<MODE, true, EB, EB_G>(zimg, ztxt, t, bias, out, g_out, gt_out,
-                                  b, n, d, ldg, diag, bx * BM, by * BN, smem);
+  tile_body<MODE, true, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
+                                       gt_out, b, n, d, ldg, diag, bx * BM,
+                                       by * BN, smem);
 }
 
 // General kernel: interior blocks take the DMA fast path, edge blocks the
 // guarded register-staged path.  MODE 0: forward loss.  MODE 1: backward
 // g-slab + scalar partials.
-template <int MODE, int EB, int EB_G = EB>
+template <int MODE, int EB, int EB_G = EB, bool NTG = false>
 __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
@@ -556,35 +565,34 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
   const bool interior = (row_base + BM <= b) && (col_base + BN <= n) &&
       (d % (128 / EB) == 0) && (n % 8 == 0);
   if (interior)
-    tile_body<MODE, true, EB, EB_G>(zimg, ztxt, t, bias, out, g_out, gt_out,
-                                    b, n, d, ldg, diag, row_base, col_base,
-                                    smem);
+    tile_body<MODE, true, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
+                                         gt_out, b, n, d, ldg, diag,
+                                         row_base, col_base, smem);
   else
-    tile_body<MODE, false, EB, EB_G>(zimg, ztxt, t, bias, out, g_out,
-                                     gt_out, b, n, d, ldg, diag, row_base,
-                                     col_base, smem);
+    tile_body<MODE, false, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
+                                          gt_out, b, n, d, ldg, diag,
+                                          row_base, col_base, smem);
 }
 
 inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
-template <int MODE, int EB, int EB_G = EB>
-int launch(uintptr_t stream, const void* zimg, const void* ztxt,
-           const void* t_prime, const void* bias, void* out, void* g_out,
-           void* gt_out, int b, int n, int d, int ldg, int diag, int flags) {
-  if (b <= 0 || n <= 0) return (int)hipSuccess;
-  if (d % (16 / EB) != 0 || ldg < n) return (int)hipErrorInvalidValue;
+template <int MODE, int EB, int EB_G = EB, bool NTG = false>
+int launch_nt(uintptr_t stream, const void* zimg, const void* ztxt,
+              const void* t_prime, const void* bias, void* out, void* g_out,
+              void* gt_out, int b, int n, int d, int ldg, int diag,
+              int flags) {
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
   const bool interior = (b % BM == 0) && (n % BN == 0) &&
       (d % (128 / EB) == 0);
   if (interior)
-    hipLaunchKernelGGL((siglip_tile_kernel_interior<MODE, EB, EB_G>), grid,
-                       dim3(THREADS), 0, (hipStream_t)stream,
+    hipLaunchKernelGGL((siglip_tile_kernel_interior<MODE, EB, EB_G, NTG>),
+                       grid, dim3(THREADS), 0, (hipStream_t)stream,
                        (const char*)zimg, (const char*)ztxt,
                        (const float*)t_prime, (const float*)bias,
                        (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
                        b, n, d, ldg, diag, flags);
   else
-    hipLaunchKernelGGL((siglip_tile_kernel<MODE, EB, EB_G>), grid,
+    hipLaunchKernelGGL((siglip_tile_kernel<MODE, EB, EB_G, NTG>), grid,
                        dim3(THREADS),
                        0, (hipStream_t)stream,
                        (const char*)zimg, (const char*)ztxt,
@@ -592,6 +600,21 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
                        (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
                        b, n, d, ldg, diag, flags);
   return (int)hipGetLastError();
+}
+
+template <int MODE, int EB, int EB_G = EB>
+int launch(uintptr_t stream, const void* zimg, const void* ztxt,
+           const void* t_prime, const void* bias, void* out, void* g_out,
+           void* gt_out, int b, int n, int d, int ldg, int diag, int flags) {
+  if (b <= 0 || n <= 0) return (int)hipSuccess;
+  if (d % (16 / EB) != 0 || ldg < n) return (int)hipErrorInvalidValue;
+  if (MODE != 0 && (flags & 4))
+    return launch_nt<MODE, EB, EB_G, true>(stream, zimg, ztxt, t_prime, bias,
+                                           out, g_out, gt_out, b, n, d, ldg,
+                                           diag, flags);
+  return launch_nt<MODE, EB, EB_G, false>(stream, zimg, ztxt, t_prime, bias,
+                                          out, g_out, gt_out, b, n, d, ldg,
+                                          diag, flags);
 }
 
 }  // namespace
