@@ -50,14 +50,18 @@ def reduce_out3(buf: torch.Tensor) -> torch.Tensor:
     return buf[:, :3].sum(dim=0)
 
 
-def _kernel_flags() -> int:
+def _kernel_flags(default_gm: str = "4") -> int:
     # bit 0: XCD-contiguous block remap — default ON: together with the
     #        grouped walk it measured best (+4%, profiles/flag_matrix_gm4.log
     #        and profiles/README.md); SIGLIP_XCD_SWZ=0 disables.
     # bit 1: grouped block walk for L2 panel reuse (default on;
     #        SIGLIP_GROUP_SWZ=0 disables for A/B profiling).
     # bit 2: non-temporal g/gᵀ slab stores (SIGLIP_NT_G=1; default off —
-    #        measure before enabling).
+    #        measured a regression on every mode, profiles round 2).
+    # bits 4-5: GROUP_M locality-group height.  Per-mode defaults from the
+    #        round-2 sweep (gpurun sweep2): plain fwd is fastest at gm=8
+    #        (854 TF), the g-emitting modes at gm=4 (the slab stores change
+    #        the L2 picture); SIGLIP_GROUP_M overrides both.
     f = 0
     if os.environ.get("SIGLIP_XCD_SWZ", "1") != "0":
         f |= 1
@@ -65,7 +69,7 @@ def _kernel_flags() -> int:
         f |= 2
     if os.environ.get("SIGLIP_NT_G", "0") == "1":
         f |= 4
-    gm = os.environ.get("SIGLIP_GROUP_M", "4")
+    gm = os.environ.get("SIGLIP_GROUP_M", default_gm)
     f |= {"8": 0, "1": 1, "4": 2, "16": 3}.get(gm, 0) << 4
     return f
 
@@ -202,7 +206,8 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         ctypes.c_void_p(stream),
         ctypes.c_void_p(zi_ptr), ctypes.c_void_p(zt_ptr),
         ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
-        ctypes.c_void_p(buf.data_ptr()), b, n, d, diag, _kernel_flags()),
+        ctypes.c_void_p(buf.data_ptr()), b, n, d, diag,
+        _kernel_flags(default_gm="8")),
         "siglip_fwd")
     return buf[:, 0].sum()
 

@@ -71,6 +71,7 @@
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) int i32x8;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 typedef const __attribute__((address_space(1))) unsigned int* gas_ptr;
 typedef __attribute__((address_space(3))) unsigned int* las_ptr;
 
@@ -397,6 +398,17 @@ __device__ __forceinline__ void tile_body(
     const unsigned lane_off_t =
         (unsigned)(wcol + (lane & 15)) * (unsigned)b
         + (unsigned)(wrow + (lane >> 4) * 4);
+    // gᵀ staging (EB_G == 1, interior, b 16-aligned): the natural gᵀ store
+    // is 4 B per lane at stride-b addresses — 16 different cache lines per
+    // quarter-wave, measured as a ~0.6 ms store tail at B=32k.  Instead
+    // stage the block's 256×256-byte gᵀ tile in LDS (the operand buffers
+    // are dead once every wave reaches its epilogue) with a 2-bit row-seg
+    // XOR swizzle, then write it back as coalesced dwordx4 columns.
+    const bool gt_lds = (EB_G == 1) && INTERIOR && ((b & 15) == 0);
+    unsigned* gt_img = reinterpret_cast<unsigned*>(smem);
+    const int gsel = (wcol + (lane & 15)) & 3;
+    const int gt_base0 = (wcol + (lane & 15)) * 64 + (lane >> 4);
+    if (gt_lds) __syncthreads();   // operand LDS reads complete everywhere
 #pragma unroll
     for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
@@ -434,7 +446,10 @@ __device__ __forceinline__ void tile_body(
         if (EB_G == 1) {
           const unsigned toff = lane_off_t + (unsigned)(ni * 16) * (unsigned)b
               + (unsigned)(mi * 16);
-          if (INTERIOR) {
+          if (gt_lds) {
+            gt_img[gt_base0 + ni * 1024 +
+                   ((((wrow >> 4) + mi) ^ gsel) << 2)] = packed;
+          } else if (INTERIOR) {
             st_g<NTG>(reinterpret_cast<unsigned*>(gtb + toff), packed);
           } else {
             const int grow0 = row_base + wrow + mi * 16 + (lane >> 4) * 4;
@@ -448,6 +463,22 @@ __device__ __forceinline__ void tile_body(
         }
       }
       __builtin_amdgcn_sched_barrier(0);  // cap epilogue register pressure
+    }
+    if (gt_lds) {
+      // Coalesced writeback: 4096 16-B chunks (16 rows of one gᵀ column
+      // each); consecutive threads cover consecutive chunks, dwordx4
+      // global stores.
+      __syncthreads();
+#pragma unroll
+      for (int sseg = 0; sseg < 8; ++sseg) {
+        const int c = sseg * THREADS + threadIdx.x;
+        const int col = c >> 4;
+        const int rowseg = c & 15;
+        const u32x4 v = *reinterpret_cast<const u32x4*>(
+            gt_img + col * 64 + ((rowseg ^ (col & 3)) << 2));
+        st_g<NTG>(reinterpret_cast<u32x4*>(
+                      gtb + (size_t)col * b + rowseg * 16), v);
+      }
     }
   }
 
