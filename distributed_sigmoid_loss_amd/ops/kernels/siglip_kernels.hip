@@ -203,7 +203,8 @@ __device__ __forceinline__ void mma_ktile_fp8(const char* smem,
 
 // g-slab element bytes follow the compute dtype: bf16 kernels emit bf16 g,
 // fp8 kernels emit e4m3 g (×448 fixed scale).
-template <int MODE, bool INTERIOR, int EB, int EB_G = EB, bool NTG = false>
+template <int MODE, bool INTERIOR, int EB, int EB_G = EB, bool NTG = false,
+          bool GT_ALIGNED = false>
 __device__ __forceinline__ void tile_body(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     float t, float bias, float* __restrict__ out, __bf16* __restrict__ g_out,
@@ -404,7 +405,10 @@ __device__ __forceinline__ void tile_body(
     // stage the block's 256×256-byte gᵀ tile in LDS (the operand buffers
     // are dead once every wave reaches its epilogue) with a 2-bit row-seg
     // XOR swizzle, then write it back as coalesced dwordx4 columns.
-    const bool gt_lds = (EB_G == 1) && INTERIOR && ((b & 15) == 0);
+    // GT_ALIGNED (b 16-aligned, checked by the caller) folds the choice at
+    // compile time so the dead store path costs no registers — keeping both
+    // paths live measured as an 80 B/lane spill.
+    constexpr bool gt_lds = (EB_G == 1) && INTERIOR && GT_ALIGNED;
     unsigned* gt_img = reinterpret_cast<unsigned*>(smem);
     const int gsel = (wcol + (lane & 15)) & 3;
     const int gt_base0 = (wcol + (lane & 15)) * 64 + (lane >> 4);
@@ -569,9 +573,9 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
   remap_block(flags, bx, by);
   const float t = __expf(*t_prime);
   const float bias = *bias_p;
-  tile_body<MODE, true, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
-                                       gt_out, b, n, d, ldg, diag, bx * BM,
-                                       by * BN, smem);
+  tile_body<MODE, true, EB, EB_G, NTG, true>(zimg, ztxt, t, bias, out,
+                                             g_out, gt_out, b, n, d, ldg,
+                                             diag, bx * BM, by * BN, smem);
 }
 
 // General kernel: interior blocks take the DMA fast path, edge blocks the
@@ -595,11 +599,17 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
 
   const bool interior = (row_base + BM <= b) && (col_base + BN <= n) &&
       (d % (128 / EB) == 0) && (n % 8 == 0);
-  if (interior)
-    tile_body<MODE, true, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
-                                         gt_out, b, n, d, ldg, diag,
-                                         row_base, col_base, smem);
-  else
+  if (interior) {
+    if (EB_G == 1 && (b & 15) == 0)
+      tile_body<MODE, true, EB, EB_G, NTG, true>(zimg, ztxt, t, bias, out,
+                                                 g_out, gt_out, b, n, d, ldg,
+                                                 diag, row_base, col_base,
+                                                 smem);
+    else
+      tile_body<MODE, true, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
+                                           gt_out, b, n, d, ldg, diag,
+                                           row_base, col_base, smem);
+  } else
     tile_body<MODE, false, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
                                           gt_out, b, n, d, ldg, diag,
                                           row_base, col_base, smem);
