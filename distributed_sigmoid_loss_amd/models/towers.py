@@ -42,7 +42,8 @@ class ProjectionTower(nn.Module):
         self.proj = nn.Linear(in_dim, emb_dim, bias=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.normalize(self.proj(x), dim=-1)
+        from ..ops import l2_normalize
+        return l2_normalize(self.proj(x))
 
 
 class TwoTowerModel(nn.Module):

@@ -110,6 +110,11 @@ def _load():
         [ctypes.c_void_p] * 8 + [ctypes.c_int] * 6)
     lib.siglip_fwdg_fp8.restype = ctypes.c_int
     lib.siglip_fwdg_fp8.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 6
+    lib.l2norm_fwd_bf16.restype = ctypes.c_int
+    lib.l2norm_fwd_bf16.argtypes = (
+        [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 + [ctypes.c_float])
+    lib.l2norm_bwd_bf16.restype = ctypes.c_int
+    lib.l2norm_bwd_bf16.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_int] * 2
     if lib.siglip_ext_abi() != 7:
         _lib_err = (f"stale HIP extension at {SO_PATH} "
                     f"(ABI {lib.siglip_ext_abi()}, need 7); rebuild with: "
@@ -499,6 +504,51 @@ def siglip_fwd_g(zimg: torch.Tensor, ztxt: torch.Tensor,
                 b, n, d, ldg, diag, _kernel_flags())
     _check(rc, "siglip_fwdg")
     return out3, g_slab, gt_slab if fp8g else None
+
+
+class _FusedL2Normalize(torch.autograd.Function):
+    """Row-wise L2 normalize via the single-pass HIP kernels (bf16, fp32
+    math; torch ``F.normalize(dim=-1)`` semantics) — replaces the ~10-kernel
+    autograd chain on the towers' hot path."""
+
+    @staticmethod
+    def forward(ctx, x, eps):
+        lib = _require_lib()
+        x = x.contiguous()
+        b, d = x.shape
+        y = torch.empty_like(x)
+        rn = torch.empty(b, device=x.device, dtype=torch.float32)
+        stream = torch.cuda.current_stream(x.device).cuda_stream
+        _check(lib.l2norm_fwd_bf16(
+            ctypes.c_void_p(stream), ctypes.c_void_p(x.data_ptr()),
+            ctypes.c_void_p(y.data_ptr()), ctypes.c_void_p(rn.data_ptr()),
+            b, d, ctypes.c_float(eps)), "l2norm_fwd")
+        ctx.save_for_backward(y, rn)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        lib = _require_lib()
+        y, rn = ctx.saved_tensors
+        dy = dy.contiguous()
+        b, d = y.shape
+        dx = torch.empty_like(y)
+        stream = torch.cuda.current_stream(y.device).cuda_stream
+        _check(lib.l2norm_bwd_bf16(
+            ctypes.c_void_p(stream), ctypes.c_void_p(dy.data_ptr()),
+            ctypes.c_void_p(y.data_ptr()), ctypes.c_void_p(rn.data_ptr()),
+            ctypes.c_void_p(dx.data_ptr()), b, d), "l2norm_bwd")
+        return dx, None
+
+
+def l2_normalize(x: torch.Tensor, eps: float = 1e-12) -> torch.Tensor:
+    """L2-normalize rows.  GPU bf16 2-D inputs take the fused kernel pair;
+    everything else falls back to ``F.normalize`` (identical semantics)."""
+    if (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 2
+            and x.shape[1] % 2 == 0 and extension_available()):
+        return _FusedL2Normalize.apply(x, eps)
+    import torch.nn.functional as F
+    return F.normalize(x, dim=-1, eps=eps)
 
 
 def siglip_bwd_from_g(zimg: torch.Tensor, ztxt: torch.Tensor,

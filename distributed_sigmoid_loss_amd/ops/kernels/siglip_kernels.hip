@@ -617,6 +617,75 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
 
 inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
+// ---------------------------------------------------------------------------
+// Fused row-wise L2 normalize (bf16 in/out, fp32 math) — replaces the
+// F.normalize autograd chain in the towers (measured 0.29 ms fwd+bwd per
+// tower at (32768, 768) as ~10 stock elementwise/reduce kernels; these two
+// single-pass kernels do the same work at memory speed).
+//   fwd: y = x / max(‖x‖₂, eps)  (torch F.normalize semantics), saves
+//        rn = 1/max(‖x‖₂, eps)
+//   bwd: dx = rn · (dy − y · ⟨y, dy⟩)   per row
+// One wave per row, 8 rows per 512-thread block, bf16-pair loads.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+
+__launch_bounds__(512) __global__ void l2norm_fwd_kernel(
+    const __bf16* __restrict__ x, __bf16* __restrict__ y,
+    float* __restrict__ rn, int b, int d, float eps) {
+  const int row = blockIdx.x * 8 + (threadIdx.x >> 6);
+  if (row >= b) return;
+  const int lane = threadIdx.x & 63;
+  const bf16x2* xr = reinterpret_cast<const bf16x2*>(x + (size_t)row * d);
+  const int dp = d >> 1;
+  float s = 0.f;
+  for (int k = lane; k < dp; k += 64) {
+    const bf16x2 v = xr[k];
+    const float a0 = (float)v.x, a1 = (float)v.y;
+    s += a0 * a0 + a1 * a1;
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) s += __shfl_down(s, off);
+  const float r = 1.0f / fmaxf(sqrtf(__shfl(s, 0)), eps);
+  bf16x2* yr = reinterpret_cast<bf16x2*>(y + (size_t)row * d);
+  for (int k = lane; k < dp; k += 64) {
+    const bf16x2 v = xr[k];
+    bf16x2 o;
+    o.x = (__bf16)((float)v.x * r);
+    o.y = (__bf16)((float)v.y * r);
+    yr[k] = o;
+  }
+  if (lane == 0) rn[row] = r;
+}
+
+__launch_bounds__(512) __global__ void l2norm_bwd_kernel(
+    const __bf16* __restrict__ dy, const __bf16* __restrict__ y,
+    const float* __restrict__ rn, __bf16* __restrict__ dx, int b, int d) {
+  const int row = blockIdx.x * 8 + (threadIdx.x >> 6);
+  if (row >= b) return;
+  const int lane = threadIdx.x & 63;
+  const bf16x2* dyr = reinterpret_cast<const bf16x2*>(dy + (size_t)row * d);
+  const bf16x2* yr = reinterpret_cast<const bf16x2*>(y + (size_t)row * d);
+  const int dp = d >> 1;
+  float dot = 0.f;
+  for (int k = lane; k < dp; k += 64) {
+    const bf16x2 a = dyr[k], c = yr[k];
+    dot += (float)a.x * (float)c.x + (float)a.y * (float)c.y;
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) dot += __shfl_down(dot, off);
+  dot = __shfl(dot, 0);
+  const float r = rn[row];
+  bf16x2* dxr = reinterpret_cast<bf16x2*>(dx + (size_t)row * d);
+  for (int k = lane; k < dp; k += 64) {
+    const bf16x2 a = dyr[k], c = yr[k];
+    bf16x2 o;
+    o.x = (__bf16)(r * ((float)a.x - (float)c.x * dot));
+    o.y = (__bf16)(r * ((float)a.y - (float)c.y * dot));
+    dxr[k] = o;
+  }
+}
+
 template <int MODE, int EB, int EB_G = EB, bool NTG = false>
 int launch_nt(uintptr_t stream, const void* zimg, const void* ztxt,
               const void* t_prime, const void* bias, void* out, void* g_out,
@@ -663,6 +732,26 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
 extern "C" {
 
 int siglip_ext_abi(void) { return 7; }
+
+int l2norm_fwd_bf16(uintptr_t stream, const void* x, void* y, void* rn,
+                    int b, int d, float eps) {
+  if (b <= 0) return (int)hipSuccess;
+  if (d <= 0 || (d & 1)) return (int)hipErrorInvalidValue;
+  hipLaunchKernelGGL(l2norm_fwd_kernel, dim3(ceil_div(b, 8)), dim3(512), 0,
+                     (hipStream_t)stream, (const __bf16*)x, (__bf16*)y,
+                     (float*)rn, b, d, eps);
+  return (int)hipGetLastError();
+}
+
+int l2norm_bwd_bf16(uintptr_t stream, const void* dy, const void* y,
+                    const void* rn, void* dx, int b, int d) {
+  if (b <= 0) return (int)hipSuccess;
+  if (d <= 0 || (d & 1)) return (int)hipErrorInvalidValue;
+  hipLaunchKernelGGL(l2norm_bwd_kernel, dim3(ceil_div(b, 8)), dim3(512), 0,
+                     (hipStream_t)stream, (const __bf16*)dy,
+                     (const __bf16*)y, (const float*)rn, (__bf16*)dx, b, d);
+  return (int)hipGetLastError();
+}
 
 int siglip_fwd_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                     const void* t_prime, const void* bias, void* loss_out,

@@ -173,6 +173,26 @@ def test_extension_is_intree_so():
         assert any("_siglip_hip.so" in line for line in f)
 
 
+@pytest.mark.parametrize("b,d", [(256, 768), (1000, 66), (8, 2)])
+def test_fused_l2_normalize(b, d):
+    """Fused single-pass normalize kernels vs F.normalize autograd chain."""
+    g = torch.Generator().manual_seed(b)
+    x32 = torch.randn(b, d, generator=g) * 3.0
+    x = x32.cuda().bfloat16().requires_grad_(True)
+    y = ops.l2_normalize(x)
+    go = torch.randn(b, d, generator=g).cuda().bfloat16()
+    y.backward(go)
+
+    x2 = x.detach().clone().requires_grad_(True)
+    y2 = F.normalize(x2.float(), dim=-1)
+    y2.backward(go.float())
+    torch.cuda.synchronize()
+
+    assert torch.allclose(y.float(), y2.detach(), rtol=1e-2, atol=1e-2)
+    assert torch.allclose(x.grad.float(), x2.grad, rtol=5e-2, atol=1e-2), \
+        (x.grad.float() - x2.grad).abs().max()
+
+
 def test_siglip_loss_module_gpu():
     """SigLipLoss (caller-owned params, reference rwightman API) through the
     fused kernels on one GPU vs CPU fp32."""
