@@ -24,6 +24,7 @@ materialized tensor on the hot path.
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
@@ -40,6 +41,7 @@ from ..parallel.ring import (
     neighbour_exchange_with_grad,
     neighbour_exchange_bidir_with_grad,
     neighbour_exchange_start,
+    neighbour_exchange_bidir_start,
 )
 from ..utils.profiling import roctx_range
 
@@ -74,7 +76,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, zimg, ztxt, t_prime, bias, group, col_chunk, impl,
-                quant, want_grad):
+                quant, want_grad, bidir=False):
         from .. import ops as _ops
         world, rank = _world_and_rank(group)
         zimg = zimg.contiguous()
@@ -145,16 +147,58 @@ class _RingAllGatherLoss(torch.autograd.Function):
                 g_slab=g_slab, gt_slab=gt_slab, col0=src * b_txt, out3=out3)
             return None   # shared out3 accumulates across chunks
 
+        use_bidir = bidir and world > 2
         if world > 1:
             left = (rank - 1 + world) % world
             right = (rank + 1) % world
             # Post hop 1 before any compute: its wire time hides under the
-            # local-block kernel below.
-            handle = neighbour_exchange_start(left, right, ztxt, group=group)
+            # local-block kernel below.  The bidirectional variant drives
+            # TWO xGMI point-to-point links per hop (reference
+            # distributed_utils.py:30-62 at the perf layer), halving the hop
+            # count to ⌈(W−1)/2⌉ — at N=8 per-rank b=4096 the ring is
+            # wire-bound (~42 µs/hop vs ~40 µs chunk compute), so hop count
+            # is the lever.
+            if use_bidir:
+                handle = neighbour_exchange_bidir_start(left, right, ztxt,
+                                                        ztxt, group=group)
+            else:
+                handle = neighbour_exchange_start(left, right, ztxt,
+                                                  group=group)
 
         loss = chunk_fwd(ztxt, rank, 0)
 
-        if world > 1:
+        def add_part(part):
+            nonlocal loss
+            if part is not None:
+                loss = loss + part
+
+        if world > 1 and use_bidir:
+            nb, rem = divmod(world - 1, 2)
+            for r in range(1, nb + 1):
+                with roctx_range(f"ring_bhop{r}_wait"):
+                    from_right, from_left = handle.wait()
+                src_r = (rank + r) % world
+                src_l = (rank - r + world) % world
+                chunks[src_r] = from_right
+                chunks[src_l] = from_left
+                if r < nb:
+                    handle = neighbour_exchange_bidir_start(
+                        left, right, from_right, from_left, group=group)
+                elif rem:
+                    # one unidirectional remainder hop: forward the
+                    # rightward-traveling stream (last received from left)
+                    handle = neighbour_exchange_start(left, right, from_left,
+                                                      group=group)
+                with roctx_range(f"ring_bchunk{r}_loss"):
+                    add_part(chunk_fwd(from_right, src_r, None))
+                    add_part(chunk_fwd(from_left, src_l, None))
+            if rem:
+                with roctx_range("ring_rem_wait"):
+                    recv = handle.wait()[0]
+                src = (rank - nb - 1 + world) % world
+                chunks[src] = recv
+                add_part(chunk_fwd(recv, src, None))
+        elif world > 1:
             for hop in range(1, world):
                 # Post hop k+1 before computing on hop k's data.  roctx
                 # ranges label the hops for rocprofv3/torch.profiler traces
@@ -167,9 +211,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
                     handle = neighbour_exchange_start(left, right, recv,
                                                       group=group)
                 with roctx_range(f"ring_chunk{hop}_loss"):
-                    part = chunk_fwd(recv, src, None)
-                    if part is not None:
-                        loss = loss + part
+                    add_part(chunk_fwd(recv, src, None))
 
         if save_g:
             # Reduce the per-XCD scalar buffers once, after every chunk's
@@ -341,7 +383,8 @@ class _RingAllGatherLoss(torch.autograd.Function):
         else:
             dztxt = dtxt_flat
 
-        return (dzimg, dztxt, dt_prime, dbias, None, None, None, None, None)
+        return (dzimg, dztxt, dt_prime, dbias, None, None, None, None, None,
+                None)
 
 
 class DistributedSigmoidLoss(nn.Module):
@@ -369,7 +412,7 @@ class DistributedSigmoidLoss(nn.Module):
 
     def __init__(self, gpu_batch_size: int, strategy: str = "all_gather",
                  col_chunk: Optional[int] = None, impl: str = "auto",
-                 quant: str = "bf16"):
+                 quant: str = "bf16", bidir: Optional[bool] = None):
         super().__init__()
         self.t_prime = nn.Parameter(torch.tensor(math.log(10.0)))
         self.bias = nn.Parameter(torch.tensor(-10.0))
@@ -382,6 +425,12 @@ class DistributedSigmoidLoss(nn.Module):
         self.col_chunk = col_chunk
         self.impl = impl
         self.quant = quant
+        # Ring hop-halving via two simultaneous xGMI links (W > 2 only).
+        # Default on; SIGLIP_RING_BIDIR=0 (or bidir=False) forces the
+        # unidirectional ring for A/B comparison.
+        if bidir is None:
+            bidir = os.environ.get("SIGLIP_RING_BIDIR", "1") != "0"
+        self.bidir = bidir
 
     def forward(self, image_embeddings: torch.Tensor,
                 text_embeddings: torch.Tensor, group=None) -> torch.Tensor:
@@ -399,7 +448,8 @@ class DistributedSigmoidLoss(nn.Module):
                                           self.t_prime, self.bias))
             total = _RingAllGatherLoss.apply(
                 image_embeddings, text_embeddings, self.t_prime, self.bias,
-                group, self.col_chunk, self.impl, self.quant, want_grad)
+                group, self.col_chunk, self.impl, self.quant, want_grad,
+                self.bidir)
         return total / self.gpu_batch_size
 
 

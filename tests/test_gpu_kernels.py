@@ -189,8 +189,9 @@ def test_fused_l2_normalize(b, d):
     torch.cuda.synchronize()
 
     assert torch.allclose(y.float(), y2.detach(), rtol=1e-2, atol=1e-2)
-    assert torch.allclose(x.grad.float(), x2.grad, rtol=5e-2, atol=1e-2), \
-        (x.grad.float() - x2.grad).abs().max()
+    assert torch.allclose(x.grad.float(), x2.grad.float(), rtol=5e-2,
+                          atol=1e-2), \
+        (x.grad.float() - x2.grad.float()).abs().max()
 
 
 def test_siglip_loss_module_gpu():

@@ -18,7 +18,9 @@ from helpers import encode_shard, run_distributed
 def ddp_step(rank, world, batch_per_rank, emb_dim, strategy, average):
     img_enc, txt_enc, zi, zt = encode_shard(rank, world, batch_per_rank,
                                             emb_dim)
-    loss_mod = DistributedSigmoidLoss(batch_per_rank, strategy=strategy)
+    bidir = strategy == "ring_bidir"
+    loss_mod = DistributedSigmoidLoss(
+        batch_per_rank, strategy="ring" if bidir else strategy, bidir=bidir)
     loss = loss_mod(zi, zt)
     loss.backward()
     if average:
@@ -67,3 +69,25 @@ def test_ring_matches_all_gather_raw_grads(world):
     gather = run_distributed(ddp_step, world, 4, 16, "all_gather", False)[0]
     for key in ("img", "txt", "t_prime", "bias", "loss"):
         assert torch.allclose(ring[key], gather[key], rtol=1e-4, atol=1e-7), key
+
+
+@pytest.mark.parametrize("world", [3, 4])
+def test_bidir_ring_matches_all_gather_raw_grads(world):
+    """The hop-halved bidirectional ring (W=3: one bidir round; W=4: bidir
+    round + unidirectional remainder hop) is numerically interchangeable
+    with the all-gather strategy."""
+    ring = run_distributed(ddp_step, world, 4, 16, "ring_bidir", False)[0]
+    gather = run_distributed(ddp_step, world, 4, 16, "all_gather", False)[0]
+    for key in ("img", "txt", "t_prime", "bias", "loss"):
+        assert torch.allclose(ring[key], gather[key], rtol=1e-4, atol=1e-7), key
+
+
+@pytest.mark.parametrize("world,batch", [(3, 3), (4, 2)])
+def test_bidir_ring_scaling_oracle(world, batch):
+    """N-rank bidir-ring grads == 1-rank grads (the reference's main
+    correctness property, test_distributed_sigmoid_loss.py:122-141)."""
+    multi = run_distributed(ddp_step, world, batch, 16, "ring_bidir", True)[0]
+    single = run_distributed(ddp_step, 1, world * batch, 16, "all_gather",
+                             True)[0]
+    for key in ("img", "txt", "t_prime", "bias"):
+        assert torch.allclose(multi[key], single[key], rtol=1e-3, atol=1e-6), key
