@@ -32,7 +32,7 @@ from distributed_sigmoid_loss_amd import DistributedSigmoidLoss
 from distributed_sigmoid_loss_amd.models import TwoTowerModel
 from distributed_sigmoid_loss_amd.parallel import average_gradients
 from distributed_sigmoid_loss_amd.utils import init_from_env, set_seed
-from distributed_sigmoid_loss_amd.utils.profiling import PhaseTimer
+from distributed_sigmoid_loss_amd.utils.profiling import PhaseTimer, HopStats
 
 
 def parse_args():
@@ -95,6 +95,8 @@ def main():
     opt = torch.optim.SGD(params, lr=1e-4)
     timer = PhaseTimer(enabled=args.csv is not None,
                        use_cuda=(device == "cuda"))
+    if args.csv and device == "cuda":
+        HopStats.set_enabled(True)   # per-hop ring comm/compute breakdown
 
     img_feats = torch.randn(b, args.dim, device=device, dtype=dtype)
     txt_feats = torch.randn(b, args.dim, device=device, dtype=dtype)
@@ -159,6 +161,10 @@ def main():
 
     if args.csv and rank == 0:
         timer.write_csv(args.csv)
+        hops = HopStats.summary()
+        if hops:
+            print("# hop_stats_ms " + json.dumps(
+                {k: round(v, 4) for k, v in hops.items()}), file=sys.stderr)
 
     ms_per_step = elapsed / args.steps * 1000.0
     pairs_per_sec = args.global_batch / (elapsed / args.steps)

@@ -43,7 +43,18 @@ from ..parallel.ring import (
     neighbour_exchange_start,
     neighbour_exchange_bidir_start,
 )
-from ..utils.profiling import roctx_range
+from ..utils.profiling import roctx_range, HopStats
+
+
+import contextlib
+
+
+@contextlib.contextmanager
+def _hop_span(label):
+    # roctx marker for external profilers + CUDA-event pair for the
+    # in-process per-hop stats (SIGLIP_HOP_STATS=1 / bench --csv).
+    with roctx_range(label), HopStats.record(label):
+        yield
 
 
 def _world_and_rank(group=None):
@@ -175,7 +186,7 @@ class _RingAllGatherLoss(torch.autograd.Function):
         if world > 1 and use_bidir:
             nb, rem = divmod(world - 1, 2)
             for r in range(1, nb + 1):
-                with roctx_range(f"ring_bhop{r}_wait"):
+                with _hop_span(f"ring_bhop{r}_wait"):
                     from_right, from_left = handle.wait()
                 src_r = (rank + r) % world
                 src_l = (rank - r + world) % world
@@ -189,11 +200,11 @@ class _RingAllGatherLoss(torch.autograd.Function):
                     # rightward-traveling stream (last received from left)
                     handle = neighbour_exchange_start(left, right, from_left,
                                                       group=group)
-                with roctx_range(f"ring_bchunk{r}_loss"):
+                with _hop_span(f"ring_bchunk{r}_loss"):
                     add_part(chunk_fwd(from_right, src_r, None))
                     add_part(chunk_fwd(from_left, src_l, None))
             if rem:
-                with roctx_range("ring_rem_wait"):
+                with _hop_span("ring_rem_wait"):
                     recv = handle.wait()[0]
                 src = (rank - nb - 1 + world) % world
                 chunks[src] = recv
@@ -203,14 +214,14 @@ class _RingAllGatherLoss(torch.autograd.Function):
                 # Post hop k+1 before computing on hop k's data.  roctx
                 # ranges label the hops for rocprofv3/torch.profiler traces
                 # (per-hop comm visibility).
-                with roctx_range(f"ring_hop{hop}_wait"):
+                with _hop_span(f"ring_hop{hop}_wait"):
                     recv = handle.wait()[0]
                 src = (rank - hop + world) % world
                 chunks[src] = recv
                 if hop < world - 1:
                     handle = neighbour_exchange_start(left, right, recv,
                                                       group=group)
-                with roctx_range(f"ring_chunk{hop}_loss"):
+                with _hop_span(f"ring_chunk{hop}_loss"):
                     add_part(chunk_fwd(recv, src, None))
 
         if save_g:

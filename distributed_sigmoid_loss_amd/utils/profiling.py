@@ -34,6 +34,50 @@ def roctx_range(name: str):
         yield
 
 
+class HopStats:
+    """Global per-hop timing collector for the ring strategy.
+
+    Enabled via SIGLIP_HOP_STATS=1 (bench --csv turns it on): the ring
+    forward wraps each exchange wait and each chunk kernel in CUDA events;
+    :func:`summary` syncs once and returns mean ms per label — the per-hop
+    comm/compute visibility SURVEY §5 asks for, without an external
+    profiler.
+    """
+
+    enabled = os.environ.get("SIGLIP_HOP_STATS", "0") == "1"
+    _events: List = []
+
+    @classmethod
+    def set_enabled(cls, on: bool):
+        cls.enabled = on
+
+    @classmethod
+    @contextlib.contextmanager
+    def record(cls, label: str):
+        if not cls.enabled or not torch.cuda.is_available():
+            yield
+            return
+        s = torch.cuda.Event(enable_timing=True)
+        e = torch.cuda.Event(enable_timing=True)
+        s.record()
+        try:
+            yield
+        finally:
+            e.record()
+            cls._events.append((label, s, e))
+
+    @classmethod
+    def summary(cls) -> Dict[str, float]:
+        if not cls._events:
+            return {}
+        torch.cuda.synchronize()
+        agg: Dict[str, List[float]] = {}
+        for n, s, e in cls._events:
+            agg.setdefault(n, []).append(s.elapsed_time(e))
+        cls._events = []
+        return {k: sum(v) / len(v) for k, v in sorted(agg.items())}
+
+
 class PhaseTimer:
     """Per-phase wall timing via CUDA events (or perf_counter on CPU).
 
