@@ -82,6 +82,28 @@ def main():
         print(f"gT@zi raw / +scale+cast: {t_a:6.3f} / {t_a2:6.3f} ms")
         print(f"g@zt  raw / +scale+cast: {t_b:6.3f} / {t_b2:6.3f} ms")
 
+    # 5b. fp8 backward pieces
+    if quant == "fp8":
+        zi_d, zt_d = zi.detach(), zt.detach()
+        qc = ops.quantize_fp8_pair(zi_d, zt_d)
+        t_q = time_fn(lambda: ops.quantize_fp8_pair(zi_d, zt_d))
+        print(f"quantize pair         : {t_q:7.3f} ms")
+        buf, g8, gt8 = ops.siglip_fwd_g(zi_d, zt_d,
+                                        loss_mod.t_prime.detach(),
+                                        loss_mod.bias.detach(), 0,
+                                        quant="fp8", qcache=qc)
+        out3 = ops.reduce_out3(buf)
+        sc = torch.tensor(0.005, device=dev)
+        t_m1 = time_fn(lambda: ops.scaled_mm8(gt8, qc[0], sc))
+        t_m2 = time_fn(lambda: ops.scaled_mm8(g8, qc[2], sc))
+        print(f"fp8 GEMM gt8@zi_q     : {t_m1:7.3f} ms")
+        print(f"fp8 GEMM g8@zt_q      : {t_m2:7.3f} ms")
+        go = torch.tensor(1.0, device=dev)
+        t_bfg8 = time_fn(lambda: ops.siglip_bwd_from_g(
+            zi_d, zt_d, loss_mod.t_prime.detach(), loss_mod.bias.detach(),
+            go, out3, g8, gt8, quant="fp8", qcache=qc))
+        print(f"bwd_from_g fp8 total  : {t_bfg8:7.3f} ms")
+
     # 6. normalize fwd+bwd alone
     x = torch.randn(b, d, device=dev, dtype=torch.bfloat16,
                     requires_grad=True)
