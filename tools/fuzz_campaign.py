@@ -78,16 +78,75 @@ def main():
             tol = 1.2e-1 if quant == "fp8" else 3e-2
             if max(errs) > tol:
                 ok = False
+
+            # saved-g path (MODE 2: fwd+g kernel + GEMM-only backward),
+            # including the 'mixed' policy on bf16-eligible cases.
+            quant2 = "mixed" if (quant == "bf16" and i % 2 == 0) else quant
+            errs2 = []
+            fl2 = 0.0
+            if quant2 == "bf16" or b % 4 == 0:
+                qc = (ops.quantize_fp8_pair(zi, zt)
+                      if quant2 in ("fp8", "mixed") else None)
+                buf, gsl, gtsl = ops.siglip_fwd_g(zi, zt, tp, bs, diag,
+                                                  quant=quant2, qcache=qc)
+                out3 = ops.reduce_out3(buf)
+                dg = ops.siglip_bwd_from_g(zi, zt, tp, bs, go, out3, gsl,
+                                           gtsl, quant=quant2, qcache=qc)
+                if quant2 == "mixed":
+                    want2 = _torch_loss(zi.float(), zt.float(), tp.float(),
+                                        bs.float(), diag, None)
+                    dr2 = _torch_bwd(zi.float(), zt.float(), tp.float(),
+                                     bs.float(), diag, go.float(), None)
+                else:
+                    want2, dr2 = want, dr
+                torch.cuda.synchronize()
+                fl2 = abs(out3[0].item() - want2.item()) / max(
+                    abs(want2.item()), 1e-6)
+                if fl2 > 3e-2:
+                    ok = False
+                errs2 = [rel_l2(dg[k], dr2[k]) for k in range(4)]
+                tol2 = 1.2e-1 if quant2 in ("fp8", "mixed") else 3e-2
+                if max(errs2) > tol2:
+                    ok = False
+
             if not ok:
                 failures += 1
-                print(f"FAIL case {i}: q={quant} b={b} n={n} d={d} "
-                      f"diag={diag} col={col} floss={fl:.3e} errs="
-                      f"{['%.3e' % e for e in errs]}")
+                print(f"FAIL case {i}: q={quant}/{quant2} b={b} n={n} d={d} "
+                      f"diag={diag} col={col} floss={fl:.3e}/{fl2:.3e} errs="
+                      f"{['%.3e' % e for e in errs]} errs2="
+                      f"{['%.3e' % e for e in errs2]}")
         except Exception as e:
             failures += 1
             print(f"ERROR case {i}: q={quant} b={b} n={n} d={d} diag={diag} "
                   f"col={col}: {type(e).__name__}: {str(e)[:160]}")
-    print(f"{n_cases - failures}/{n_cases} passed")
+    # fused L2-normalize fuzz (fwd+bwd vs fp32 F.normalize)
+    n_norm = max(10, n_cases // 5)
+    for i in range(n_norm):
+        b = rng.randrange(1, 5000)
+        d = 2 * rng.randrange(1, 1200)
+        g = torch.Generator().manual_seed(rng.randrange(1 << 30))
+        scale = rng.uniform(0.05, 5.0)
+        x = (torch.randn(b, d, generator=g) * scale).cuda().bfloat16()
+        x = x.clone().requires_grad_(True)
+        gy = torch.randn(b, d, generator=g).cuda().bfloat16()
+        try:
+            y = ops.l2_normalize(x)
+            y.backward(gy)
+            x2 = x.detach().float().clone().requires_grad_(True)
+            y2 = F.normalize(x2, dim=-1)
+            y2.backward(gy.float())
+            torch.cuda.synchronize()
+            e_y = rel_l2(y.detach(), y2.detach())
+            e_g = rel_l2(x.grad, x2.grad)
+            if e_y > 2e-2 or e_g > 3e-2:
+                failures += 1
+                print(f"NORM FAIL case {i}: b={b} d={d} scale={scale:.2f} "
+                      f"e_y={e_y:.3e} e_g={e_g:.3e}")
+        except Exception as e:
+            failures += 1
+            print(f"NORM ERROR case {i}: b={b} d={d}: "
+                  f"{type(e).__name__}: {str(e)[:160]}")
+    print(f"{n_cases + n_norm - failures}/{n_cases + n_norm} passed")
     sys.exit(1 if failures else 0)
 
 
