@@ -412,6 +412,11 @@ __device__ __forceinline__ void tile_body(
     unsigned* gt_img = reinterpret_cast<unsigned*>(smem);
     const int gsel = (wcol + (lane & 15)) & 3;
     const int gt_base0 = (wcol + (lane & 15)) * 64 + (lane >> 4);
+    // The fp8 g slab gets the same treatment (1-B row-major scatter →
+    // 64-KB LDS tile at smem+64K, byte writes merge per dword, dwordx4
+    // row writeback); together with gt_img this uses the whole 128 KB.
+    unsigned char* g_img = reinterpret_cast<unsigned char*>(smem) + 65536;
+    const int g_base0 = (wrow + (lane >> 4) * 4) * 256 + wcol + (lane & 15);
     if (gt_lds) __syncthreads();   // operand LDS reads complete everywhere
 #pragma unroll
     for (int mi = 0; mi < FM; ++mi) {
@@ -436,8 +441,11 @@ __device__ __forceinline__ void tile_body(
             const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)ldg;
             if (EB_G == 1) {
               const unsigned char q = __hip_fp8_e4m3(g * 448.0f).__x;
-              st_g<NTG>(reinterpret_cast<unsigned char*>(gb)
-                            + lane_off + row_off + ni * 16, q);
+              if (gt_lds)
+                g_img[g_base0 + (mi * 16 + reg) * 256 + ni * 16] = q;
+              else
+                st_g<NTG>(reinterpret_cast<unsigned char*>(gb)
+                              + lane_off + row_off + ni * 16, q);
               packed |= (unsigned)q << (8 * reg);
             } else {
               st_g<NTG>(reinterpret_cast<__bf16*>(gb)
@@ -469,8 +477,9 @@ __device__ __forceinline__ void tile_body(
       __builtin_amdgcn_sched_barrier(0);  // cap epilogue register pressure
     }
     if (gt_lds) {
-      // Coalesced writeback: 4096 16-B chunks (16 rows of one gᵀ column
-      // each); consecutive threads cover consecutive chunks, dwordx4
+      // Coalesced writeback: 4096 16-B chunks per slab.  gᵀ: 16 rows of
+      // one column each (swizzled); g: 16 columns of one row (row-major,
+      // direct).  Consecutive threads cover consecutive chunks, dwordx4
       // global stores.
       __syncthreads();
 #pragma unroll
@@ -482,6 +491,10 @@ __device__ __forceinline__ void tile_body(
             gt_img + col * 64 + ((rowseg ^ (col & 3)) << 2));
         st_g<NTG>(reinterpret_cast<u32x4*>(
                       gtb + (size_t)col * b + rowseg * 16), v);
+        const u32x4 w = *reinterpret_cast<const u32x4*>(
+            g_img + col * 256 + rowseg * 16);
+        st_g<NTG>(reinterpret_cast<u32x4*>(
+                      gb + (size_t)col * ldg + rowseg * 16), w);
       }
     }
   }
@@ -600,7 +613,7 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
   const bool interior = (row_base + BM <= b) && (col_base + BN <= n) &&
       (d % (128 / EB) == 0) && (n % 8 == 0);
   if (interior) {
-    if (EB_G == 1 && (b & 15) == 0)
+    if (EB_G == 1 && (b & 15) == 0 && (ldg & 15) == 0)
       tile_body<MODE, true, EB, EB_G, NTG, true>(zimg, ztxt, t, bias, out,
                                                  g_out, gt_out, b, n, d, ldg,
                                                  diag, row_base, col_base,
@@ -692,8 +705,11 @@ int launch_nt(uintptr_t stream, const void* zimg, const void* ztxt,
               void* gt_out, int b, int n, int d, int ldg, int diag,
               int flags) {
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
+  // The interior-only kernel folds the LDS-staged g/gᵀ writeback at compile
+  // time (GT_ALIGNED), which needs 16-aligned slab strides for its dwordx4
+  // stores — odd ldg shapes take the general kernel's guarded dispatch.
   const bool interior = (b % BM == 0) && (n % BN == 0) &&
-      (d % (128 / EB) == 0);
+      (d % (128 / EB) == 0) && (EB_G != 1 || (ldg & 15) == 0);
   if (interior)
     hipLaunchKernelGGL((siglip_tile_kernel_interior<MODE, EB, EB_G, NTG>),
                        grid, dim3(THREADS), 0, (hipStream_t)stream,
