@@ -417,7 +417,14 @@ __device__ __forceinline__ void tile_body(
     // row writeback); together with gt_img this uses the whole 128 KB.
     unsigned char* g_img = reinterpret_cast<unsigned char*>(smem) + 65536;
     const int g_base0 = (wrow + (lane >> 4) * 4) * 256 + wcol + (lane & 15);
-    if (gt_lds) __syncthreads();   // operand LDS reads complete everywhere
+    // bf16 g staging: the 256×256-element bf16 tile is exactly the whole
+    // 128-KB smem (no gᵀ slab exists in bf16 mode).  Row-quad XOR swizzle
+    // (((row>>2)&7)<<5, ≥ bit 5) keeps 16-B chunks intact and spreads the
+    // 2-B scatter writes across all banks.
+    constexpr bool g_lds16 = (EB_G == 2) && INTERIOR && GT_ALIGNED;
+    char* g_img16 = smem;
+    const int g16_base0 = (wcol + (lane & 15)) * 2;
+    if (gt_lds || g_lds16) __syncthreads();   // operand LDS reads complete
 #pragma unroll
     for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
@@ -447,6 +454,11 @@ __device__ __forceinline__ void tile_body(
                 st_g<NTG>(reinterpret_cast<unsigned char*>(gb)
                               + lane_off + row_off + ni * 16, q);
               packed |= (unsigned)q << (8 * reg);
+            } else if (g_lds16) {
+              const int rloc = wrow + mi * 16 + (lane >> 4) * 4 + reg;
+              *reinterpret_cast<__bf16*>(g_img16 + rloc * 512 +
+                  ((g16_base0 + ni * 32) ^ (((rloc >> 2) & 7) << 5))) =
+                  (__bf16)g;
             } else {
               st_g<NTG>(reinterpret_cast<__bf16*>(gb)
                             + lane_off + row_off + ni * 16, (__bf16)g);
@@ -476,6 +488,21 @@ __device__ __forceinline__ void tile_body(
       }
       __builtin_amdgcn_sched_barrier(0);  // cap epilogue register pressure
     }
+    if (g_lds16) {
+      // bf16 writeback: 8192 16-B chunks (8 elements of one row each),
+      // nt dwordx4; the XOR swizzle is inverted per (row, chunk).
+      __syncthreads();
+#pragma unroll
+      for (int sseg = 0; sseg < 16; ++sseg) {
+        const int c = sseg * THREADS + threadIdx.x;
+        const int row = c >> 5;
+        const int cs = c & 31;
+        const u32x4 v = *reinterpret_cast<const u32x4*>(
+            g_img16 + row * 512 + ((cs * 16) ^ (((row >> 2) & 7) << 5)));
+        st_g<true>(reinterpret_cast<u32x4*>(
+                       gb + (size_t)row * (ldg * EB_G) + cs * 16), v);
+      }
+    }
     if (gt_lds) {
       // Coalesced writeback: 4096 16-B chunks per slab.  gᵀ: 16 rows of
       // one column each (swizzled); g: 16 columns of one row (row-major,
@@ -489,12 +516,12 @@ __device__ __forceinline__ void tile_body(
         const int rowseg = c & 15;
         const u32x4 v = *reinterpret_cast<const u32x4*>(
             gt_img + col * 64 + ((rowseg ^ (col & 3)) << 2));
-        st_g<NTG>(reinterpret_cast<u32x4*>(
-                      gtb + (size_t)col * b + rowseg * 16), v);
+        st_g<true>(reinterpret_cast<u32x4*>(
+                       gtb + (size_t)col * b + rowseg * 16), v);
         const u32x4 w = *reinterpret_cast<const u32x4*>(
             g_img + col * 256 + rowseg * 16);
-        st_g<NTG>(reinterpret_cast<u32x4*>(
-                      gb + (size_t)col * ldg + rowseg * 16), w);
+        st_g<true>(reinterpret_cast<u32x4*>(
+                       gb + (size_t)col * ldg + rowseg * 16), w);
       }
     }
   }
@@ -613,7 +640,8 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
   const bool interior = (row_base + BM <= b) && (col_base + BN <= n) &&
       (d % (128 / EB) == 0) && (n % 8 == 0);
   if (interior) {
-    if (EB_G == 1 && (b & 15) == 0 && (ldg & 15) == 0)
+    if ((EB_G == 1 && (b & 15) == 0 && (ldg & 15) == 0) ||
+        (EB_G == 2 && MODE != 0 && (ldg & 7) == 0))
       tile_body<MODE, true, EB, EB_G, NTG, true>(zimg, ztxt, t, bias, out,
                                                  g_out, gt_out, b, n, d, ldg,
                                                  diag, row_base, col_base,
@@ -709,7 +737,8 @@ int launch_nt(uintptr_t stream, const void* zimg, const void* ztxt,
   // time (GT_ALIGNED), which needs 16-aligned slab strides for its dwordx4
   // stores — odd ldg shapes take the general kernel's guarded dispatch.
   const bool interior = (b % BM == 0) && (n % BN == 0) &&
-      (d % (128 / EB) == 0) && (EB_G != 1 || (ldg & 15) == 0);
+      (d % (128 / EB) == 0) &&
+      (EB_G == 1 ? (ldg & 15) == 0 : (MODE == 0 || (ldg & 7) == 0));
   if (interior)
     hipLaunchKernelGGL((siglip_tile_kernel_interior<MODE, EB, EB_G, NTG>),
                        grid, dim3(THREADS), 0, (hipStream_t)stream,
