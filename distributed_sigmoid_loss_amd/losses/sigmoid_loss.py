@@ -442,6 +442,9 @@ class DistributedSigmoidLoss(nn.Module):
         if bidir is None:
             bidir = os.environ.get("SIGLIP_RING_BIDIR", "1") != "0"
         self.bidir = bidir
+        if quant in ("fp8", "mixed") and torch.cuda.is_available():
+            from .. import ops as _ops
+            _ops.load_tuned_gemms()
 
     def forward(self, image_embeddings: torch.Tensor,
                 text_embeddings: torch.Tensor, group=None) -> torch.Tensor:
