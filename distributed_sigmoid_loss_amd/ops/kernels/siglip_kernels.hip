@@ -776,7 +776,65 @@ int launch(uintptr_t stream, const void* zimg, const void* ztxt,
 
 extern "C" {
 
-int siglip_ext_abi(void) { return 7; }
+int siglip_ext_abi(void) { return 8; }
+
+// Fused per-tensor fp8-e4m3 quantization: one amax pass (block reduce +
+// one atomicMax of the float bits per block — positive floats order as
+// uints) and one cast pass reading the amax from device memory (no host
+// sync).  Replaces ~5 stock kernels / ~0.25 ms per quantize-pair at B=32k.
+
+__launch_bounds__(256) __global__ void amax_abs_bf16_kernel(
+    const __bf16* __restrict__ x, unsigned* __restrict__ amax_bits,
+    long long n) {
+  float m = 0.f;
+  const long long stride = (long long)gridDim.x * 256 * 8;
+  for (long long i = ((long long)blockIdx.x * 256 + threadIdx.x) * 8;
+       i < n; i += stride) {
+    if (i + 8 <= n) {
+      const bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) m = fmaxf(m, fabsf((float)v[k]));
+    } else {
+      for (long long j = i; j < n; ++j) m = fmaxf(m, fabsf((float)x[j]));
+    }
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_down(m, off));
+  __shared__ float red[4];
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = m;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    m = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+    atomicMax(amax_bits, __float_as_uint(m));
+  }
+}
+
+__launch_bounds__(256) __global__ void quant_fp8_bf16_kernel(
+    const __bf16* __restrict__ x, unsigned char* __restrict__ q,
+    float* __restrict__ scale_out, const unsigned* __restrict__ amax_bits,
+    long long n) {
+  const float amax = fmaxf(__uint_as_float(*amax_bits), 9.5367431640625e-7f);
+  const float r = 448.0f / amax;
+  if (blockIdx.x == 0 && threadIdx.x == 0) *scale_out = amax / 448.0f;
+  const long long stride = (long long)gridDim.x * 256 * 8;
+  for (long long i = ((long long)blockIdx.x * 256 + threadIdx.x) * 8;
+       i < n; i += stride) {
+    if (i + 8 <= n) {
+      const bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i);
+      unsigned lo = 0, hi = 0;
+#pragma unroll
+      for (int k = 0; k < 4; ++k)
+        lo |= (unsigned)__hip_fp8_e4m3((float)v[k] * r).__x << (8 * k);
+#pragma unroll
+      for (int k = 0; k < 4; ++k)
+        hi |= (unsigned)__hip_fp8_e4m3((float)v[4 + k] * r).__x << (8 * k);
+      *reinterpret_cast<uint2*>(q + i) = {lo, hi};
+    } else {
+      for (long long j = i; j < n; ++j)
+        q[j] = __hip_fp8_e4m3((float)x[j] * r).__x;
+    }
+  }
+}
 
 int l2norm_fwd_bf16(uintptr_t stream, const void* x, void* y, void* rn,
                     int b, int d, float eps) {
@@ -795,6 +853,24 @@ int l2norm_bwd_bf16(uintptr_t stream, const void* dy, const void* y,
   hipLaunchKernelGGL(l2norm_bwd_kernel, dim3(ceil_div(b, 8)), dim3(512), 0,
                      (hipStream_t)stream, (const __bf16*)dy,
                      (const __bf16*)y, (const float*)rn, (__bf16*)dx, b, d);
+  return (int)hipGetLastError();
+}
+
+// Per-tensor e4m3 quantization.  amax_bits must be a zeroed 4-B buffer;
+// q gets n e4m3 bytes, scale_out the fp32 scale (x ≈ q · scale).
+// n must be a multiple of 8 for the vectorized path to cover all of it
+// exactly when (x, q) are 16-B aligned; any n works (tail loop).
+int quant_fp8_bf16(uintptr_t stream, const void* x, void* q,
+                   void* scale_out, void* amax_bits, long long n) {
+  if (n <= 0) return (int)hipSuccess;
+  const int blocks = 2048;
+  hipLaunchKernelGGL(amax_abs_bf16_kernel, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, (const __bf16*)x,
+                     (unsigned*)amax_bits, n);
+  hipLaunchKernelGGL(quant_fp8_bf16_kernel, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, (const __bf16*)x,
+                     (unsigned char*)q, (float*)scale_out,
+                     (const unsigned*)amax_bits, n);
   return (int)hipGetLastError();
 }
 

@@ -172,6 +172,24 @@ def test_fwdg_quant_vs_fp32(quant):
     assert torch.allclose(tp.grad.float(), tp2.grad, rtol=rtol, atol=atol)
 
 
+@pytest.mark.parametrize("b,d", [(256, 768), (1000, 120), (7, 8)])
+def test_fused_quant_matches_torch(b, d):
+    """Fused amax+cast quantization kernels vs the torch composite."""
+    g = torch.Generator().manual_seed(b + d)
+    x = (torch.randn(b, d, generator=g) * 2.5).cuda().bfloat16()
+    q, s = ops._quant_fp8(x)
+    # torch composite reference
+    amax = x.detach().abs().amax().float().clamp_(min=2.0 ** -20)
+    s_ref = amax / 448.0
+    q_ref = (x.float() / s_ref).to(torch.float8_e4m3fn)
+    torch.cuda.synchronize()
+    assert torch.allclose(s.cpu(), s_ref.cpu(), rtol=1e-6)
+    # identical scale => elementwise cast should agree except for ties
+    diff = (q.float() - q_ref.float()).abs()
+    denom = q_ref.float().abs().clamp(min=1.0)
+    assert (diff / denom).max().item() < 0.07, (diff / denom).max()
+
+
 def test_inference_skips_slab():
     """Under no_grad the plain forward kernel runs (no g allocation) — the
     loss is identical either way."""
