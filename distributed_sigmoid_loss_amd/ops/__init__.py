@@ -149,7 +149,7 @@ def _load():
     lib.l2norm_bwd_bf16.restype = ctypes.c_int
     lib.l2norm_bwd_bf16.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_int] * 2
     lib.quant_fp8_bf16.restype = ctypes.c_int
-    lib.quant_fp8_bf16.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_longlong]
+    lib.quant_fp8_bf16.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_longlong]
     _lib = lib
     return _lib
 
