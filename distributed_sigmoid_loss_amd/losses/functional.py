@@ -169,13 +169,22 @@ class _FusedSigmoidLoss(torch.autograd.Function):
         from .. import ops
         zimg = zimg.contiguous()
         ztxt = ztxt.contiguous()
-        qc = (ops.quantize_fp8_pair(zimg, ztxt)
-              if (quant in ("fp8", "mixed") and zimg.is_cuda) else None)
         b, n = zimg.shape[0], ztxt.shape[0]
+        d = zimg.shape[1]
         # want_grad is computed by the CALLER (grad mode is disabled inside
         # Function.forward, so it cannot be probed here).
         save_g = (want_grad and col_chunk is None and zimg.is_cuda
                   and ops.save_g_enabled(b, n, quant))
+        qc = None
+        if quant in ("fp8", "mixed") and zimg.is_cuda:
+            # fp8 + saved-g on aligned shapes: ROW-wise pow2 scales, hardware
+            # dequant in the MX MFMA (exact per-row dynamic range); otherwise
+            # per-tensor scales folded into the temperature.
+            if (quant == "fp8" and save_g and ops.extension_available()
+                    and ops.rowwise_ok(b, n, d)):
+                qc = ops.quantize_fp8_rowwise_pair(zimg, ztxt)
+            else:
+                qc = ops.quantize_fp8_pair(zimg, ztxt)
         if save_g:
             buf, g, gt = ops.siglip_fwd_g(zimg, ztxt, t_prime, bias,
                                           diag_offset, quant=quant,

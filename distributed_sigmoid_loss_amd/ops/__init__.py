@@ -120,9 +120,9 @@ def _load():
         _lib_err = f"failed to load {SO_PATH}: {e}"
         return None
     lib.siglip_ext_abi.restype = ctypes.c_int
-    if lib.siglip_ext_abi() != 8:
+    if lib.siglip_ext_abi() != 9:
         _lib_err = (f"stale HIP extension at {SO_PATH} "
-                    f"(ABI {lib.siglip_ext_abi()}, need 8); rebuild with: "
+                    f"(ABI {lib.siglip_ext_abi()}, need 9); rebuild with: "
                     "python -m distributed_sigmoid_loss_amd.ops.build --force")
         return None
     lib.siglip_fwd_bf16.restype = ctypes.c_int
@@ -130,9 +130,13 @@ def _load():
     lib.siglip_bwd_g_bf16.restype = ctypes.c_int
     lib.siglip_bwd_g_bf16.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] * 5
     lib.siglip_fwd_fp8.restype = ctypes.c_int
-    lib.siglip_fwd_fp8.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] * 5
+    lib.siglip_fwd_fp8.argtypes = ([ctypes.c_void_p] * 6
+                                   + [ctypes.c_int] * 5
+                                   + [ctypes.c_void_p] * 2)
     lib.siglip_bwd_g_fp8.restype = ctypes.c_int
-    lib.siglip_bwd_g_fp8.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5
+    lib.siglip_bwd_g_fp8.argtypes = ([ctypes.c_void_p] * 8
+                                     + [ctypes.c_int] * 5
+                                     + [ctypes.c_void_p] * 4)
     lib.siglip_bwd_g_mixed.restype = ctypes.c_int
     lib.siglip_bwd_g_mixed.argtypes = (
         [ctypes.c_void_p] * 8 + [ctypes.c_int] * 5)
@@ -142,7 +146,9 @@ def _load():
     lib.siglip_fwdg_mixed.argtypes = (
         [ctypes.c_void_p] * 8 + [ctypes.c_int] * 6)
     lib.siglip_fwdg_fp8.restype = ctypes.c_int
-    lib.siglip_fwdg_fp8.argtypes = [ctypes.c_void_p] * 8 + [ctypes.c_int] * 6
+    lib.siglip_fwdg_fp8.argtypes = ([ctypes.c_void_p] * 8
+                                    + [ctypes.c_int] * 6
+                                    + [ctypes.c_void_p] * 4)
     lib.l2norm_fwd_bf16.restype = ctypes.c_int
     lib.l2norm_fwd_bf16.argtypes = (
         [ctypes.c_void_p] * 4 + [ctypes.c_int] * 2 + [ctypes.c_float])
@@ -150,6 +156,9 @@ def _load():
     lib.l2norm_bwd_bf16.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_int] * 2
     lib.quant_fp8_bf16.restype = ctypes.c_int
     lib.quant_fp8_bf16.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_longlong]
+    lib.quant_fp8_rowwise_bf16.restype = ctypes.c_int
+    lib.quant_fp8_rowwise_bf16.argtypes = (
+        [ctypes.c_void_p] * 5 + [ctypes.c_int] * 2)
     _lib = lib
     return _lib
 
@@ -220,6 +229,45 @@ def _quant_fp8(x: torch.Tensor):
     return q, scale
 
 
+def _quant_fp8_rowwise(x: torch.Tensor):
+    """Per-ROW pow2 quantization for the hardware-scaled MX MFMA path.
+
+    Returns (q, e8, ratio, s_ref): x_row ≈ q_row · 2^(e8_row−127);
+    ratio_row = 2^(e8_row − e_max) ≤ 1 (the backward slab fold), and
+    s_ref = 2^(e_max−127) (the per-tensor factor the GEMM scale carries).
+    """
+    lib = _require_lib()
+    xd = x.detach().contiguous()
+    b, d = xd.shape
+    q = torch.empty((b, d), device=xd.device, dtype=torch.float8_e4m3fn)
+    e8 = torch.empty(b, device=xd.device, dtype=torch.uint8)
+    emax = torch.zeros(1, device=xd.device, dtype=torch.int32)
+    stream = torch.cuda.current_stream(xd.device).cuda_stream
+    _check(lib.quant_fp8_rowwise_bf16(
+        ctypes.c_void_p(stream), ctypes.c_void_p(xd.data_ptr()),
+        ctypes.c_void_p(q.data_ptr()), ctypes.c_void_p(e8.data_ptr()),
+        ctypes.c_void_p(emax.data_ptr()), b, d), "quant_fp8_rowwise")
+    emax_f = emax[0].float()
+    ratio = torch.exp2(e8.float() - emax_f)
+    s_ref = torch.exp2(emax_f - 127.0)
+    return q, e8, ratio, s_ref
+
+
+def rowwise_ok(b: int, n: int, d: int) -> bool:
+    """Row-wise fp8 policy gate: needs the saved-g mm8-aligned shapes (the
+    slab folds assume the fp8 GEMM path) and SIGLIP_FP8_ROWWISE != 0."""
+    return (os.environ.get("SIGLIP_FP8_ROWWISE", "1") != "0"
+            and b % 16 == 0 and n % 16 == 0 and d % 16 == 0
+            and hasattr(torch, "_scaled_mm"))
+
+
+def quantize_fp8_rowwise_pair(zimg: torch.Tensor, ztxt: torch.Tensor):
+    """Row-wise qcache: an 8-tuple (zi_q, zi_e8, zi_ratio, si_ref,
+    zt_q, zt_e8, zt_ratio, st_ref) — distinguished from the per-tensor
+    4-tuple by length."""
+    return _quant_fp8_rowwise(zimg) + _quant_fp8_rowwise(ztxt)
+
+
 def quantize_fp8_pair(zimg: torch.Tensor, ztxt: torch.Tensor):
     """Quantize both embedding tensors once; pass the result as ``qcache`` to
     both :func:`siglip_fwd` and :func:`siglip_bwd` so an fwd+bwd step pays a
@@ -241,10 +289,18 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     dev = zimg.device
     tp = _prep_scalar(t_prime, dev)
     bp = _prep_scalar(bias, dev)
+    extra = ()
     if quant == "fp8":
-        zi_q, si, zt_q, st = (qcache if qcache is not None
-                              else quantize_fp8_pair(zimg, ztxt))
-        tp = tp + si.log() + st.log()
+        if qcache is not None and len(qcache) == 8:
+            # row-wise policy: hardware dequant, tp stays unmodified
+            zi_q, zi_e8, _, _, zt_q, zt_e8, _, _ = qcache
+            extra = (ctypes.c_void_p(zi_e8.data_ptr()),
+                     ctypes.c_void_p(zt_e8.data_ptr()))
+        else:
+            zi_q, si, zt_q, st = (qcache if qcache is not None
+                                  else quantize_fp8_pair(zimg, ztxt))
+            tp = tp + si.log() + st.log()
+            extra = (None, None)
         zi_ptr, zt_ptr = zi_q.data_ptr(), zt_q.data_ptr()
         fn = lib.siglip_fwd_fp8
     else:
@@ -258,7 +314,7 @@ def siglip_fwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         ctypes.c_void_p(zi_ptr), ctypes.c_void_p(zt_ptr),
         ctypes.c_void_p(tp.data_ptr()), ctypes.c_void_p(bp.data_ptr()),
         ctypes.c_void_p(buf.data_ptr()), b, n, d, diag,
-        _kernel_flags(default_gm="8")),
+        _kernel_flags(default_gm="8"), *extra),
         "siglip_fwd")
     return buf[:, 0].sum()
 
@@ -288,6 +344,11 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
         # Reuse the forward's quantization when provided (identical inputs
         # give identical amax, so recomputing is equivalent but wasteful);
         # the g kernel sees t_eff so its logits match the forward's.
+        # (The recompute path is per-tensor only — row-wise qcaches ride
+        # the saved-g path, ops.siglip_bwd_from_g.)
+        if qcache is not None and len(qcache) == 8:
+            raise RuntimeError(
+                "row-wise fp8 qcache requires the saved-g backward")
         zi_q, si, zt_q, st = (qcache if qcache is not None
                               else quantize_fp8_pair(zimg, ztxt))
         zi_g, zt_g = zi_q, zt_q           # fp8 GEMM operands
@@ -352,6 +413,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
     def run_g(j0, j1, g_slab, diag, gt_slab=None):
         zt_slab = zt_k[j0:j1]
         if fp8g:
+            extra = ((None, None, None, None) if quant == "fp8" else ())
             _check(g_fn(
                 ctypes.c_void_p(stream),
                 ctypes.c_void_p(zi_k.data_ptr()),
@@ -361,7 +423,7 @@ def siglip_bwd(zimg: torch.Tensor, ztxt: torch.Tensor, t_prime: torch.Tensor,
                 ctypes.c_void_p(g_slab.data_ptr()),
                 ctypes.c_void_p(gt_slab.data_ptr()),
                 ctypes.c_void_p(scal.data_ptr()),
-                b, j1 - j0, d, diag, _kernel_flags()),
+                b, j1 - j0, d, diag, _kernel_flags(), *extra),
                 "siglip_bwd_g_fp8/mixed")
         else:
             _check(g_fn(
@@ -524,10 +586,22 @@ def siglip_fwd_g(zimg: torch.Tensor, ztxt: torch.Tensor,
             "recompute path (save_g_enabled would have said no)")
     if out3 is None:
         out3 = _out_buf(dev)
+    rs_extra = ()
     if quant == "fp8":
-        zi_q, si, zt_q, st = (qcache if qcache is not None
-                              else quantize_fp8_pair(zimg, ztxt))
-        tp_k = tp + si.log() + st.log()
+        if qcache is not None and len(qcache) == 8:
+            # row-wise: e8m0 scales feed the MFMA, ratios feed the slab
+            # folds; tp stays unmodified (acc is the true dot).
+            zi_q, zi_e8, zi_rat, _, zt_q, zt_e8, zt_rat, _ = qcache
+            tp_k = tp
+            rs_extra = (ctypes.c_void_p(zi_e8.data_ptr()),
+                        ctypes.c_void_p(zt_e8.data_ptr()),
+                        ctypes.c_void_p(zi_rat.data_ptr()),
+                        ctypes.c_void_p(zt_rat.data_ptr()))
+        else:
+            zi_q, si, zt_q, st = (qcache if qcache is not None
+                                  else quantize_fp8_pair(zimg, ztxt))
+            tp_k = tp + si.log() + st.log()
+            rs_extra = (None, None, None, None)
         fn = lib.siglip_fwdg_fp8
         zi_ptr, zt_ptr = zi_q.data_ptr(), zt_q.data_ptr()
     else:
@@ -541,7 +615,8 @@ def siglip_fwd_g(zimg: torch.Tensor, ztxt: torch.Tensor,
                 ctypes.c_void_p(zt_ptr), ctypes.c_void_p(tp_k.data_ptr()),
                 ctypes.c_void_p(bp.data_ptr()),
                 ctypes.c_void_p(out3.data_ptr()), ctypes.c_void_p(g_ptr),
-                ctypes.c_void_p(gt_ptr), b, n, d, ldg, diag, _kernel_flags())
+                ctypes.c_void_p(gt_ptr), b, n, d, ldg, diag, _kernel_flags(),
+                *rs_extra)
     else:
         rc = fn(ctypes.c_void_p(stream), ctypes.c_void_p(zi_ptr),
                 ctypes.c_void_p(zt_ptr), ctypes.c_void_p(tp_k.data_ptr()),
@@ -617,7 +692,17 @@ def siglip_bwd_from_g(zimg: torch.Tensor, ztxt: torch.Tensor,
     go = grad_output.detach().reshape(()).to(device=dev, dtype=torch.float32)
     scale = go * t_true
     fp8g = quant in ("fp8", "mixed")
-    if fp8g:
+    if fp8g and qcache is not None and len(qcache) == 8:
+        # row-wise policy: the slabs already fold the per-row ratios; the
+        # GEMM scale carries only the tensor-max factor, and dt' needs no
+        # scale correction (the MFMA accumulated the true dot).
+        zi_q, _, _, si_ref, zt_q, _, _, st_ref = qcache
+        dztxt = scaled_mm8(gt, zi_q, (scale / 448.0) * si_ref)
+        if on_dztxt is not None:
+            on_dztxt(dztxt)
+        dzimg = scaled_mm8(g, zt_q, (scale / 448.0) * st_ref)
+        t_eff = t_true
+    elif fp8g:
         zi_q, si, zt_q, st = qcache
         use_mm8 = (b % 16 == 0 and n % 16 == 0 and d % 16 == 0
                    and hasattr(torch, "_scaled_mm"))

@@ -172,12 +172,18 @@ __device__ __forceinline__ void mma_half_bf16(const char* smem, int aAddr,
           afrag[mi], bfrag[ni], acc[mi][ni], 0, 0, 0);
 }
 
-// One fp8 K-step (128 elements): MX-scaled MFMA, unit block scales (e8m0 =
-// 127 → 1.0); per-tensor scaling is folded into the temperature upstream.
+// One fp8 K-step (128 elements): MX-scaled MFMA.  sA/sB are per-fragment
+// scale dwords: unit (0x7f7f7f7f) for the per-tensor policy (scales folded
+// into the temperature upstream), or per-ROW e8m0 bytes replicated across
+// the dword for the row-wise policy (lane l of a fragment holds row
+// lane&15, K-block lane>>4 — validated by tools/probe_mx_rowscale.hip) so
+// the hardware emits the exact row-scaled dot.
 __device__ __forceinline__ void mma_ktile_fp8(const char* smem,
                                               const int* aAddr,
                                               const int* bAddr,
-                                              f32x4 (&acc)[FM][FN]) {
+                                              f32x4 (&acc)[FM][FN],
+                                              const int* sA,
+                                              const int* sB) {
   uint4 blo[FN], bhi[FN];
 #pragma unroll
   for (int ni = 0; ni < FN; ++ni) {
@@ -197,18 +203,21 @@ __device__ __forceinline__ void mma_ktile_fp8(const char* smem,
     for (int ni = 0; ni < FN; ++ni)
       acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
           af, pack8(blo[ni], bhi[ni]), acc[mi][ni], 0, 0,
-          0, 0x7f7f7f7f, 0, 0x7f7f7f7f);
+          0, sA[mi], 0, sB[ni]);
   }
 }
 
 // g-slab element bytes follow the compute dtype: bf16 kernels emit bf16 g,
 // fp8 kernels emit e4m3 g (×448 fixed scale).
 template <int MODE, bool INTERIOR, int EB, int EB_G = EB, bool NTG = false,
-          bool GT_ALIGNED = false>
+          bool GT_ALIGNED = false, bool RS = false>
 __device__ __forceinline__ void tile_body(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     float t, float bias, float* __restrict__ out, __bf16* __restrict__ g_out,
     unsigned char* __restrict__ gt_out,
+    const unsigned char* __restrict__ a_e8,
+    const unsigned char* __restrict__ b_e8,
+    const float* __restrict__ a_rat, const float* __restrict__ b_rat,
     int b, int n, int d, int ldg, int diag, int row_base, int col_base,
     char* smem) {
   const int lane = threadIdx.x & 63;
@@ -235,6 +244,31 @@ __device__ __forceinline__ void tile_body(
     const int c = (EB == 2) ? (kk * 4 + qbase) : (qbase * 2 + kk);
     aAddr[kk] = chunk_addr(wrow + fr, c);
     bAddr[kk] = TILE_BYTES + chunk_addr(wcol + fr, c);
+  }
+
+  // fp8 MFMA scale dwords (per fragment, K-invariant: row-wise scales).
+  int sA[FM], sB[FN];
+  if (EB == 1) {
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi) {
+      int e = 127;
+      if (RS) {
+        const int r = row_base + wrow + mi * 16 + (lane & 15);
+        e = (INTERIOR || r < b) ? (int)a_e8[INTERIOR ? r : (r < b ? r : 0)]
+                                : 127;
+      }
+      sA[mi] = 0x01010101 * e;
+    }
+#pragma unroll
+    for (int ni = 0; ni < FN; ++ni) {
+      int e = 127;
+      if (RS) {
+        const int c = col_base + wcol + ni * 16 + (lane & 15);
+        e = (INTERIOR || c < n) ? (int)b_e8[INTERIOR ? c : (c < n ? c : 0)]
+                                : 127;
+      }
+      sB[ni] = 0x01010101 * e;
+    }
   }
 
   if (INTERIOR) {
@@ -321,7 +355,7 @@ __device__ __forceinline__ void tile_body(
         } else {
           asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
         }
-        mma_ktile_fp8(smem, aAddr, bAddr, acc);
+        mma_ktile_fp8(smem, aAddr, bAddr, acc, sA, sB);
         aAddr[0] ^= 2 * TILE_BYTES; aAddr[1] ^= 2 * TILE_BYTES;
         bAddr[0] ^= 2 * TILE_BYTES; bAddr[1] ^= 2 * TILE_BYTES;
         asm volatile("s_barrier" ::: "memory");
@@ -342,7 +376,7 @@ __device__ __forceinline__ void tile_body(
         mma_half_bf16(smem, aAddr[0], bAddr[0], acc);
         mma_half_bf16(smem, aAddr[1], bAddr[1], acc);
       } else {
-        mma_ktile_fp8(smem, aAddr, bAddr, acc);
+        mma_ktile_fp8(smem, aAddr, bAddr, acc, sA, sB);
       }
       aAddr[0] ^= 2 * TILE_BYTES; aAddr[1] ^= 2 * TILE_BYTES;
       bAddr[0] ^= 2 * TILE_BYTES; bAddr[1] ^= 2 * TILE_BYTES;
@@ -425,8 +459,31 @@ __device__ __forceinline__ void tile_body(
     char* g_img16 = smem;
     const int g16_base0 = (wcol + (lane & 15)) * 2;
     if (gt_lds || g_lds16) __syncthreads();   // operand LDS reads complete
+    // Row-wise policy: the saved slabs fold the OTHER side's scale ratio
+    // (g slab ← text-row ratio for the dzimg GEMM, gᵀ slab ← image-row
+    // ratio for the dztxt GEMM), each normalized by the tensor max scale
+    // so |g·ratio| ≤ 1 keeps the fixed ×448 e4m3 packing exact; the
+    // Python side multiplies the GEMM scale by s_ref/448.
+    float rbv[FN];
+#pragma unroll
+    for (int ni = 0; ni < FN; ++ni) {
+      rbv[ni] = 1.0f;
+      if (RS && EB_G == 1) {
+        const int c = col_base + wcol + ni * 16 + (lane & 15);
+        rbv[ni] = (INTERIOR || c < n) ? b_rat[(INTERIOR || c < n) ? c : 0]
+                                      : 0.0f;
+      }
+    }
 #pragma unroll
     for (int mi = 0; mi < FM; ++mi) {
+      float rav[4] = {1.0f, 1.0f, 1.0f, 1.0f};
+      if (RS && EB_G == 1) {
+        const int r0 = row_base + wrow + mi * 16 + (lane >> 4) * 4;
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr)
+          rav[rr] = (INTERIOR || r0 + rr < b)
+              ? a_rat[(INTERIOR || r0 + rr < b) ? (r0 + rr) : 0] : 0.0f;
+      }
 #pragma unroll
       for (int ni = 0; ni < FN; ++ni) {
         unsigned packed = 0;
@@ -447,13 +504,16 @@ __device__ __forceinline__ void tile_body(
             const float g = pos ? -sig : sig;
             const unsigned row_off = (unsigned)(mi * 16 + reg) * (unsigned)ldg;
             if (EB_G == 1) {
-              const unsigned char q = __hip_fp8_e4m3(g * 448.0f).__x;
+              const unsigned char q = __hip_fp8_e4m3(
+                  (RS ? g * rbv[ni] : g) * 448.0f).__x;
               if (gt_lds)
                 g_img[g_base0 + (mi * 16 + reg) * 256 + ni * 16] = q;
               else
                 st_g<NTG>(reinterpret_cast<unsigned char*>(gb)
                               + lane_off + row_off + ni * 16, q);
-              packed |= (unsigned)q << (8 * reg);
+              const unsigned char qt = RS
+                  ? __hip_fp8_e4m3(g * rav[reg] * 448.0f).__x : q;
+              packed |= (unsigned)qt << (8 * reg);
             } else if (g_lds16) {
               const int rloc = wrow + mi * 16 + (lane >> 4) * 4 + reg;
               *reinterpret_cast<__bf16*>(g_img16 + rloc * 512 +
@@ -601,32 +661,38 @@ __device__ __forceinline__ void remap_block(int flags, int& bx, int& by) {
 // Interior-only kernel: every tile full, d % K-step == 0, n%8==0 — checked
 // by the host launcher.  Separate from the general kernel so the hot path's
 // register allocation is not inflated by the guarded path.
-template <int MODE, int EB, int EB_G = EB, bool NTG = false>
+template <int MODE, int EB, int EB_G = EB, bool NTG = false, bool RS = false>
 __launch_bounds__(THREADS) __global__ void siglip_tile_kernel_interior(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
     float* __restrict__ out, __bf16* __restrict__ g_out,
     unsigned char* __restrict__ gt_out,
+    const unsigned char* __restrict__ a_e8,
+    const unsigned char* __restrict__ b_e8,
+    const float* __restrict__ a_rat, const float* __restrict__ b_rat,
     int b, int n, int d, int ldg, int diag, int flags) {
   __shared__ char smem[4 * TILE_BYTES];
   int bx, by;
   remap_block(flags, bx, by);
   const float t = __expf(*t_prime);
   const float bias = *bias_p;
-  tile_body<MODE, true, EB, EB_G, NTG, true>(zimg, ztxt, t, bias, out,
-                                             g_out, gt_out, b, n, d, ldg,
-                                             diag, bx * BM, by * BN, smem);
+  tile_body<MODE, true, EB, EB_G, NTG, true, RS>(
+      zimg, ztxt, t, bias, out, g_out, gt_out, a_e8, b_e8, a_rat, b_rat,
+      b, n, d, ldg, diag, bx * BM, by * BN, smem);
 }
 
 // General kernel: interior blocks take the DMA fast path, edge blocks the
 // guarded register-staged path.  MODE 0: forward loss.  MODE 1: backward
 // g-slab + scalar partials.
-template <int MODE, int EB, int EB_G = EB, bool NTG = false>
+template <int MODE, int EB, int EB_G = EB, bool NTG = false, bool RS = false>
 __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
     const char* __restrict__ zimg, const char* __restrict__ ztxt,
     const float* __restrict__ t_prime, const float* __restrict__ bias_p,
     float* __restrict__ out, __bf16* __restrict__ g_out,
     unsigned char* __restrict__ gt_out,
+    const unsigned char* __restrict__ a_e8,
+    const unsigned char* __restrict__ b_e8,
+    const float* __restrict__ a_rat, const float* __restrict__ b_rat,
     int b, int n, int d, int ldg, int diag, int flags) {
   __shared__ char smem[4 * TILE_BYTES];
   int bx, by;
@@ -642,18 +708,17 @@ __launch_bounds__(THREADS) __global__ void siglip_tile_kernel(
   if (interior) {
     if ((EB_G == 1 && (b & 15) == 0 && (ldg & 15) == 0) ||
         (EB_G == 2 && MODE != 0 && (ldg & 7) == 0))
-      tile_body<MODE, true, EB, EB_G, NTG, true>(zimg, ztxt, t, bias, out,
-                                                 g_out, gt_out, b, n, d, ldg,
-                                                 diag, row_base, col_base,
-                                                 smem);
+      tile_body<MODE, true, EB, EB_G, NTG, true, RS>(
+          zimg, ztxt, t, bias, out, g_out, gt_out, a_e8, b_e8, a_rat, b_rat,
+          b, n, d, ldg, diag, row_base, col_base, smem);
     else
-      tile_body<MODE, true, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
-                                           gt_out, b, n, d, ldg, diag,
-                                           row_base, col_base, smem);
+      tile_body<MODE, true, EB, EB_G, NTG, false, RS>(
+          zimg, ztxt, t, bias, out, g_out, gt_out, a_e8, b_e8, a_rat, b_rat,
+          b, n, d, ldg, diag, row_base, col_base, smem);
   } else
-    tile_body<MODE, false, EB, EB_G, NTG>(zimg, ztxt, t, bias, out, g_out,
-                                          gt_out, b, n, d, ldg, diag,
-                                          row_base, col_base, smem);
+    tile_body<MODE, false, EB, EB_G, NTG, false, RS>(
+        zimg, ztxt, t, bias, out, g_out, gt_out, a_e8, b_e8, a_rat, b_rat,
+        b, n, d, ldg, diag, row_base, col_base, smem);
 }
 
 inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
@@ -727,11 +792,12 @@ __launch_bounds__(512) __global__ void l2norm_bwd_kernel(
   }
 }
 
-template <int MODE, int EB, int EB_G = EB, bool NTG = false>
+template <int MODE, int EB, int EB_G = EB, bool NTG = false, bool RS = false>
 int launch_nt(uintptr_t stream, const void* zimg, const void* ztxt,
               const void* t_prime, const void* bias, void* out, void* g_out,
-              void* gt_out, int b, int n, int d, int ldg, int diag,
-              int flags) {
+              void* gt_out, const void* a_e8, const void* b_e8,
+              const void* a_rat, const void* b_rat, int b, int n, int d,
+              int ldg, int diag, int flags) {
   dim3 grid(ceil_div(b, BM), ceil_div(n, BN));
   // The interior-only kernel folds the LDS-staged g/gᵀ writeback at compile
   // time (GT_ALIGNED), which needs 16-aligned slab strides for its dwordx4
@@ -740,43 +806,57 @@ int launch_nt(uintptr_t stream, const void* zimg, const void* ztxt,
       (d % (128 / EB) == 0) &&
       (EB_G == 1 ? (ldg & 15) == 0 : (MODE == 0 || (ldg & 7) == 0));
   if (interior)
-    hipLaunchKernelGGL((siglip_tile_kernel_interior<MODE, EB, EB_G, NTG>),
-                       grid, dim3(THREADS), 0, (hipStream_t)stream,
-                       (const char*)zimg, (const char*)ztxt,
-                       (const float*)t_prime, (const float*)bias,
-                       (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
-                       b, n, d, ldg, diag, flags);
+    hipLaunchKernelGGL(
+        (siglip_tile_kernel_interior<MODE, EB, EB_G, NTG, RS>),
+        grid, dim3(THREADS), 0, (hipStream_t)stream,
+        (const char*)zimg, (const char*)ztxt,
+        (const float*)t_prime, (const float*)bias,
+        (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
+        (const unsigned char*)a_e8, (const unsigned char*)b_e8,
+        (const float*)a_rat, (const float*)b_rat,
+        b, n, d, ldg, diag, flags);
   else
-    hipLaunchKernelGGL((siglip_tile_kernel<MODE, EB, EB_G, NTG>), grid,
-                       dim3(THREADS),
-                       0, (hipStream_t)stream,
-                       (const char*)zimg, (const char*)ztxt,
-                       (const float*)t_prime, (const float*)bias,
-                       (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
-                       b, n, d, ldg, diag, flags);
+    hipLaunchKernelGGL(
+        (siglip_tile_kernel<MODE, EB, EB_G, NTG, RS>), grid, dim3(THREADS),
+        0, (hipStream_t)stream,
+        (const char*)zimg, (const char*)ztxt,
+        (const float*)t_prime, (const float*)bias,
+        (float*)out, (__bf16*)g_out, (unsigned char*)gt_out,
+        (const unsigned char*)a_e8, (const unsigned char*)b_e8,
+        (const float*)a_rat, (const float*)b_rat,
+        b, n, d, ldg, diag, flags);
   return (int)hipGetLastError();
 }
 
 template <int MODE, int EB, int EB_G = EB>
 int launch(uintptr_t stream, const void* zimg, const void* ztxt,
            const void* t_prime, const void* bias, void* out, void* g_out,
-           void* gt_out, int b, int n, int d, int ldg, int diag, int flags) {
+           void* gt_out, int b, int n, int d, int ldg, int diag, int flags,
+           const void* a_e8 = nullptr, const void* b_e8 = nullptr,
+           const void* a_rat = nullptr, const void* b_rat = nullptr) {
   if (b <= 0 || n <= 0) return (int)hipSuccess;
   if (d % (16 / EB) != 0 || ldg < n) return (int)hipErrorInvalidValue;
+  const bool rs = (EB == 1) && a_e8 != nullptr;
+  if (rs && MODE != 0 && (a_rat == nullptr || b_rat == nullptr))
+    return (int)hipErrorInvalidValue;
+  if (rs)
+    return launch_nt<MODE, EB, EB_G, false, true>(
+        stream, zimg, ztxt, t_prime, bias, out, g_out, gt_out, a_e8, b_e8,
+        a_rat, b_rat, b, n, d, ldg, diag, flags);
   if (MODE != 0 && (flags & 4))
-    return launch_nt<MODE, EB, EB_G, true>(stream, zimg, ztxt, t_prime, bias,
-                                           out, g_out, gt_out, b, n, d, ldg,
-                                           diag, flags);
-  return launch_nt<MODE, EB, EB_G, false>(stream, zimg, ztxt, t_prime, bias,
-                                          out, g_out, gt_out, b, n, d, ldg,
-                                          diag, flags);
+    return launch_nt<MODE, EB, EB_G, true>(
+        stream, zimg, ztxt, t_prime, bias, out, g_out, gt_out, a_e8, b_e8,
+        a_rat, b_rat, b, n, d, ldg, diag, flags);
+  return launch_nt<MODE, EB, EB_G, false>(
+      stream, zimg, ztxt, t_prime, bias, out, g_out, gt_out, a_e8, b_e8,
+      a_rat, b_rat, b, n, d, ldg, diag, flags);
 }
 
 }  // namespace
 
 extern "C" {
 
-int siglip_ext_abi(void) { return 8; }
+int siglip_ext_abi(void) { return 9; }
 
 // Fused per-tensor fp8-e4m3 quantization: one amax pass (block reduce +
 // one atomicMax of the float bits per block — positive floats order as
@@ -836,6 +916,49 @@ __launch_bounds__(256) __global__ void quant_fp8_bf16_kernel(
   }
 }
 
+// Row-wise pow2 (e8m0) quantization for the hardware-scaled MX path:
+// per row, k = ceil(log2(amax/448)), e8m0 byte = 127+k, q = x·2^-k (so
+// |q| ≤ 448 uses the full e4m3 range and the MFMA's 2^(e-127) dequant is
+// exact).  Also maintains the tensor-max exponent (atomicMax) for the
+// backward-fold normalization.  One wave per row.
+__launch_bounds__(512) __global__ void quant_fp8_rowwise_kernel(
+    const __bf16* __restrict__ x, unsigned char* __restrict__ q,
+    unsigned char* __restrict__ e8, int* __restrict__ emax,
+    int b, int d) {
+  const int row = blockIdx.x * 8 + (threadIdx.x >> 6);
+  if (row >= b) return;
+  const int lane = threadIdx.x & 63;
+  const __bf16* xr = x + (size_t)row * d;
+  float m = 0.f;
+  for (int k = lane; k < d; k += 64) m = fmaxf(m, fabsf((float)xr[k]));
+#pragma unroll
+  for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_down(m, off));
+  m = __shfl(m, 0);
+  const float t = fmaxf(m, 1e-30f) / 448.0f;
+  const unsigned bits = __float_as_uint(t);
+  int k2 = (int)((bits >> 23) & 0xff) - 127 + ((bits & 0x7fffffu) ? 1 : 0);
+  if (k2 < -126) k2 = -126;
+  if (k2 > 127) k2 = 127;
+  const float r = __uint_as_float((unsigned)(127 - k2) << 23);   // 2^-k
+  unsigned char* qr = q + (size_t)row * d;
+  for (int k = lane; k < d; k += 64)
+    qr[k] = __hip_fp8_e4m3((float)xr[k] * r).__x;
+  if (lane == 0) {
+    e8[row] = (unsigned char)(127 + k2);
+    atomicMax(emax, 127 + k2);
+  }
+}
+
+int quant_fp8_rowwise_bf16(uintptr_t stream, const void* x, void* q,
+                           void* e8, void* emax, int b, int d) {
+  if (b <= 0 || d <= 0) return (int)hipErrorInvalidValue;
+  hipLaunchKernelGGL(quant_fp8_rowwise_kernel, dim3(ceil_div(b, 8)),
+                     dim3(512), 0, (hipStream_t)stream, (const __bf16*)x,
+                     (unsigned char*)q, (unsigned char*)e8, (int*)emax,
+                     b, d);
+  return (int)hipGetLastError();
+}
+
 int l2norm_fwd_bf16(uintptr_t stream, const void* x, void* y, void* rn,
                     int b, int d, float eps) {
   if (b <= 0) return (int)hipSuccess;
@@ -888,11 +1011,17 @@ int siglip_bwd_g_bf16(uintptr_t stream, const void* zimg, const void* ztxt,
                       nullptr, b, n, d, n, diag, flags);
 }
 
+// fp8 entry points: the trailing scale pointers select the policy —
+// null → per-tensor (scales folded into the temperature upstream);
+// non-null → row-wise e8m0 hardware dequant (a_e8/b_e8: per-row exponent
+// bytes; a_rat/b_rat: fp32 per-row scale ÷ tensor-max scale, consumed by
+// the slab folds in MODE 1/2).
 int siglip_fwd_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
                    const void* t_prime, const void* bias, void* loss_out,
-                   int b, int n, int d, int diag, int flags) {
+                   int b, int n, int d, int diag, int flags,
+                   const void* a_e8, const void* b_e8) {
   return launch<0, 1>(stream, zimg, ztxt, t_prime, bias, loss_out, nullptr,
-                      nullptr, b, n, d, n, diag, flags);
+                      nullptr, b, n, d, n, diag, flags, a_e8, b_e8);
 }
 
 // Mixed policy: bf16 logits recompute, e4m3 ×448 g and gᵀ slabs (for the
@@ -910,10 +1039,12 @@ int siglip_bwd_g_mixed(uintptr_t stream, const void* zimg, const void* ztxt,
 int siglip_bwd_g_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
                      const void* t_prime, const void* bias, void* g_out,
                      void* gt_out, void* scal, int b, int n, int d, int diag,
-                     int flags) {
+                     int flags, const void* a_e8, const void* b_e8,
+                     const void* a_rat, const void* b_rat) {
   if (b % 4 != 0) return (int)hipErrorInvalidValue;  // packed 4-B gt stores
   return launch<1, 1>(stream, zimg, ztxt, t_prime, bias, scal, g_out,
-                      gt_out, b, n, d, n, diag, flags);
+                      gt_out, b, n, d, n, diag, flags, a_e8, b_e8, a_rat,
+                      b_rat);
 }
 
 // ---- fwd+g ("saved-g") entry points: one kernel computes the loss, the g
@@ -945,10 +1076,12 @@ int siglip_fwdg_mixed(uintptr_t stream, const void* zimg, const void* ztxt,
 int siglip_fwdg_fp8(uintptr_t stream, const void* zimg, const void* ztxt,
                     const void* t_prime, const void* bias, void* out,
                     void* g_out, void* gt_out, int b, int n, int d,
-                    int ldg, int diag, int flags) {
+                    int ldg, int diag, int flags, const void* a_e8,
+                    const void* b_e8, const void* a_rat, const void* b_rat) {
   if (b % 4 != 0) return (int)hipErrorInvalidValue;
   return launch<2, 1>(stream, zimg, ztxt, t_prime, bias, out, g_out,
-                      gt_out, b, n, d, ldg, diag, flags);
+                      gt_out, b, n, d, ldg, diag, flags, a_e8, b_e8, a_rat,
+                      b_rat);
 }
 
 }  // extern "C"
