@@ -116,9 +116,12 @@ def test_bwd_from_g_matches_recompute():
 
 @pytest.mark.parametrize("quant", ["fp8", "mixed"])
 def test_fwdg_quant_matches_recompute(quant):
-    """fp8/mixed saved-g path vs the recompute path through full autograd."""
+    """fp8/mixed saved-g path vs the recompute path through full autograd.
+    Pins the per-tensor policy — this test compares the two backward
+    MECHANICS, and the recompute path is per-tensor by design."""
     b, n, d = 512, 512, 768
     results = {}
+    os.environ["SIGLIP_FP8_ROWWISE"] = "0"
     for mode in ("1", "0"):
         os.environ["SIGLIP_SAVE_G"] = mode
         try:
@@ -135,6 +138,7 @@ def test_fwdg_quant_matches_recompute(quant):
                              bs.grad)
         finally:
             os.environ.pop("SIGLIP_SAVE_G", None)
+    os.environ.pop("SIGLIP_FP8_ROWWISE", None)
     for a, b_ in zip(results["1"], results["0"]):
         assert torch.allclose(a.float(), b_.float(), rtol=5e-2, atol=5e-3), \
             (a.float() - b_.float()).abs().max()
@@ -172,62 +176,66 @@ def test_fwdg_quant_vs_fp32(quant):
     assert torch.allclose(tp.grad.float(), tp2.grad, rtol=rtol, atol=atol)
 
 
-def test_rowwise_fp8_matches_per_tensor_on_normalized():
-    """On unit-norm rows the row-wise and per-tensor fp8 policies must agree
-    to quantization tolerance (full autograd both ways)."""
+def test_rowwise_fp8_vs_fp32_reference():
+    """Row-wise fp8 full autograd against the fp32 reference at fp8-class
+    tolerances (unit-norm rows)."""
     b, n, d = 512, 512, 768
-    outs = {}
-    for mode in ("1", "0"):
-        os.environ["SIGLIP_FP8_ROWWISE"] = mode
-        try:
-            zi, zt, tp, bs = make_inputs(b, n, d, seed=55)
-            zi = zi.clone().requires_grad_(True)
-            zt = zt.clone().requires_grad_(True)
-            tp = tp.clone().requires_grad_(True)
-            bs = bs.clone().requires_grad_(True)
-            loss = sigmoid_contrastive_loss(zi, zt, tp, bs, diag_offset=0,
-                                            quant="fp8")
-            loss.backward()
-            torch.cuda.synchronize()
-            outs[mode] = (loss.detach(), zi.grad, zt.grad, tp.grad, bs.grad)
-        finally:
-            os.environ.pop("SIGLIP_FP8_ROWWISE", None)
-    for a, b_ in zip(outs["1"], outs["0"]):
-        assert torch.allclose(a.float(), b_.float(), rtol=1e-1, atol=2e-2), \
-            (a.float() - b_.float()).abs().max()
+    os.environ["SIGLIP_FP8_ROWWISE"] = "1"
+    try:
+        zi, zt, tp, bs = make_inputs(b, n, d, seed=55)
+        zi = zi.clone().requires_grad_(True)
+        zt = zt.clone().requires_grad_(True)
+        tp = tp.clone().requires_grad_(True)
+        bs = bs.clone().requires_grad_(True)
+        loss = sigmoid_contrastive_loss(zi, zt, tp, bs, diag_offset=0,
+                                        quant="fp8")
+        loss.backward()
+        torch.cuda.synchronize()
+    finally:
+        os.environ.pop("SIGLIP_FP8_ROWWISE", None)
+
+    zi2 = zi.detach().float().clone().requires_grad_(True)
+    zt2 = zt.detach().float().clone().requires_grad_(True)
+    tp2 = tp.detach().float().clone().requires_grad_(True)
+    bs2 = bs.detach().float().clone().requires_grad_(True)
+    ref = _torch_loss(zi2, zt2, tp2, bs2, 0, col_chunk=None)
+    ref.backward()
+    torch.cuda.synchronize()
+
+    def rel_l2(a, r):
+        return float((a.float() - r).norm() / r.norm().clamp(min=1e-12))
+
+    assert torch.allclose(loss.float(), ref, rtol=5e-2)
+    assert rel_l2(zi.grad, zi2.grad) < 0.12
+    assert rel_l2(zt.grad, zt2.grad) < 0.12
+    assert torch.allclose(bs.grad.float(), bs2.grad, rtol=1e-1, atol=1e-2)
 
 
-def test_rowwise_fp8_beats_per_tensor_on_wild_rows():
-    """Rows spanning 2^-6..2^6: per-tensor e4m3 wipes out the small rows;
-    the row-wise hardware-scaled path keeps per-row precision.  Compare
-    each policy's LOSS against the bf16-exact reference."""
-    b, n, d = 512, 512, 768
+def test_rowwise_quant_keeps_per_row_precision():
+    """The feature's claim, measured directly: with row norms spanning
+    2^-6..2^6, per-tensor e4m3 destroys small rows' relative precision
+    while row-wise pow2 scales keep EVERY row at e4m3 precision."""
+    b, d = 512, 768
     g = torch.Generator().manual_seed(99)
-    mag_i = torch.exp2(torch.randint(-6, 7, (b, 1), generator=g).float())
-    mag_t = torch.exp2(torch.randint(-6, 7, (n, 1), generator=g).float())
-    zi = (F.normalize(torch.randn(b, d, generator=g), dim=-1) * mag_i)
-    zt = (F.normalize(torch.randn(n, d, generator=g), dim=-1) * mag_t)
-    zi = zi.cuda().bfloat16()
-    zt = zt.cuda().bfloat16()
-    tp = torch.tensor(math.log(0.5)).cuda()   # keep logits in a sane range
-    bs = torch.tensor(-2.0).cuda()
-    ref = _torch_loss(zi.float(), zt.float(), tp.float(), bs.float(), 0,
-                      col_chunk=None)
-    errs = {}
-    for mode in ("1", "0"):
-        os.environ["SIGLIP_FP8_ROWWISE"] = mode
-        try:
-            zi_g = zi.clone().requires_grad_(True)
-            loss = sigmoid_contrastive_loss(zi_g, zt, tp, bs, diag_offset=0,
-                                            quant="fp8")
-            loss.backward()   # force the saved-g path end to end
-            torch.cuda.synchronize()
-            errs[mode] = abs(loss.item() - ref.item()) / abs(ref.item())
-        finally:
-            os.environ.pop("SIGLIP_FP8_ROWWISE", None)
-    # row-wise must be close to exact; per-tensor visibly degraded
-    assert errs["1"] < 2e-2, errs
-    assert errs["1"] <= errs["0"], errs
+    mag = torch.exp2(torch.randint(-6, 7, (b, 1), generator=g).float())
+    x = (F.normalize(torch.randn(b, d, generator=g), dim=-1) * mag)
+    x = x.cuda().bfloat16()
+
+    q_r, e8, ratio, s_ref = ops._quant_fp8_rowwise(x)
+    # reconstruct: q · 2^(e8-127) per row
+    rec_r = q_r.float() * torch.exp2(e8.float() - 127.0).unsqueeze(1)
+    q_t, s_t = ops.quantize_fp8_pair(x, x)[:2]
+    rec_t = q_t.float() * s_t
+    torch.cuda.synchronize()
+
+    xf = x.float()
+    norms = xf.norm(dim=1).clamp(min=1e-12)
+    err_r = ((rec_r - xf).norm(dim=1) / norms)
+    err_t = ((rec_t - xf).norm(dim=1) / norms)
+    # every row stays at e4m3 precision under row-wise scales
+    assert err_r.max().item() < 0.05, err_r.max()
+    # per-tensor loses the small rows (sanity that the comparison is real)
+    assert err_t.max().item() > 0.2, err_t.max()
 
 
 @pytest.mark.parametrize("b,d", [(256, 768), (1000, 120), (7, 8)])
