@@ -929,8 +929,14 @@ __launch_bounds__(512) __global__ void quant_fp8_rowwise_kernel(
   if (row >= b) return;
   const int lane = threadIdx.x & 63;
   const __bf16* xr = x + (size_t)row * d;
+  const bf16x2* xr2 = reinterpret_cast<const bf16x2*>(xr);
+  const int dp = d >> 1;
   float m = 0.f;
-  for (int k = lane; k < d; k += 64) m = fmaxf(m, fabsf((float)xr[k]));
+  for (int k = lane; k < dp; k += 64) {
+    const bf16x2 v = xr2[k];
+    m = fmaxf(m, fmaxf(fabsf((float)v.x), fabsf((float)v.y)));
+  }
+  if (d & 1) m = fmaxf(m, fabsf((float)xr[d - 1]));
 #pragma unroll
   for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_down(m, off));
   m = __shfl(m, 0);
@@ -941,8 +947,15 @@ __launch_bounds__(512) __global__ void quant_fp8_rowwise_kernel(
   if (k2 > 127) k2 = 127;
   const float r = __uint_as_float((unsigned)(127 - k2) << 23);   // 2^-k
   unsigned char* qr = q + (size_t)row * d;
-  for (int k = lane; k < d; k += 64)
-    qr[k] = __hip_fp8_e4m3((float)xr[k] * r).__x;
+  for (int k = lane; k < dp; k += 64) {
+    const bf16x2 v = xr2[k];
+    unsigned short pk =
+        (unsigned short)__hip_fp8_e4m3((float)v.x * r).__x
+        | ((unsigned short)__hip_fp8_e4m3((float)v.y * r).__x << 8);
+    reinterpret_cast<unsigned short*>(qr)[k] = pk;
+  }
+  if ((d & 1) && lane == 0)
+    qr[d - 1] = __hip_fp8_e4m3((float)xr[d - 1] * r).__x;
   if (lane == 0) {
     e8[row] = (unsigned char)(127 + k2);
     atomicMax(emax, 127 + k2);
