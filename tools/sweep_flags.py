@@ -78,6 +78,21 @@ def main():
         t = time_fn(lambda: ops.siglip_fwd_g(zi, zt, tp, bs, 0,
                                              quant="mixed", qcache=qc))
         print(f"nt={nt}              : {t:7.3f} ms  {tf/t:7.1f} TF")
+    print("-- fp8 row-wise decomposition --")
+    setenv(1, 4, 0)
+    t_qrw = time_fn(lambda: ops.quantize_fp8_rowwise_pair(zi, zt))
+    print(f"rowwise quant pair   : {t_qrw:7.3f} ms")
+    qrw = ops.quantize_fp8_rowwise_pair(zi, zt)
+    t_frw = time_fn(lambda: ops.siglip_fwd_g(zi, zt, tp, bs, 0, quant="fp8",
+                                             qcache=qrw))
+    print(f"fwd+g fp8 rowwise    : {t_frw:7.3f} ms  {tf/t_frw:7.1f} TF")
+    buf, g8, gt8 = ops.siglip_fwd_g(zi, zt, tp, bs, 0, quant="fp8",
+                                    qcache=qrw)
+    out3 = ops.reduce_out3(buf)
+    go = torch.tensor(1.0, device="cuda")
+    t_brw = time_fn(lambda: ops.siglip_bwd_from_g(
+        zi, zt, tp, bs, go, out3, g8, gt8, quant="fp8", qcache=qrw))
+    print(f"bwd_from_g rowwise   : {t_brw:7.3f} ms")
     print("-- bwd recompute g-kernel full slab (MODE1) --")
     go = torch.tensor(1.0, device="cuda")
     for nt in (0, 1):
