@@ -929,14 +929,17 @@ __launch_bounds__(512) __global__ void quant_fp8_rowwise_kernel(
   if (row >= b) return;
   const int lane = threadIdx.x & 63;
   const __bf16* xr = x + (size_t)row * d;
-  const bf16x2* xr2 = reinterpret_cast<const bf16x2*>(xr);
-  const int dp = d >> 1;
+  const bf16x8* xr8 = reinterpret_cast<const bf16x8*>(xr);
+  // vector path needs 16-B row alignment: d % 8 (rows then stay aligned)
+  const int d8 = ((d & 7) == 0) ? (d >> 3) : 0;
   float m = 0.f;
-  for (int k = lane; k < dp; k += 64) {
-    const bf16x2 v = xr2[k];
-    m = fmaxf(m, fmaxf(fabsf((float)v.x), fabsf((float)v.y)));
+  for (int k = lane; k < d8; k += 64) {
+    const bf16x8 v = xr8[k];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) m = fmaxf(m, fabsf((float)v[e]));
   }
-  if (d & 1) m = fmaxf(m, fabsf((float)xr[d - 1]));
+  for (int k = (d8 << 3) + lane; k < d; k += 64)
+    m = fmaxf(m, fabsf((float)xr[k]));
 #pragma unroll
   for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_down(m, off));
   m = __shfl(m, 0);
@@ -947,15 +950,19 @@ __launch_bounds__(512) __global__ void quant_fp8_rowwise_kernel(
   if (k2 > 127) k2 = 127;
   const float r = __uint_as_float((unsigned)(127 - k2) << 23);   // 2^-k
   unsigned char* qr = q + (size_t)row * d;
-  for (int k = lane; k < dp; k += 64) {
-    const bf16x2 v = xr2[k];
-    unsigned short pk =
-        (unsigned short)__hip_fp8_e4m3((float)v.x * r).__x
-        | ((unsigned short)__hip_fp8_e4m3((float)v.y * r).__x << 8);
-    reinterpret_cast<unsigned short*>(qr)[k] = pk;
+  for (int k = lane; k < d8; k += 64) {
+    const bf16x8 v = xr8[k];
+    unsigned lo = 0, hi = 0;
+#pragma unroll
+    for (int e = 0; e < 4; ++e)
+      lo |= (unsigned)__hip_fp8_e4m3((float)v[e] * r).__x << (8 * e);
+#pragma unroll
+    for (int e = 0; e < 4; ++e)
+      hi |= (unsigned)__hip_fp8_e4m3((float)v[4 + e] * r).__x << (8 * e);
+    reinterpret_cast<uint2*>(qr)[k] = {lo, hi};
   }
-  if ((d & 1) && lane == 0)
-    qr[d - 1] = __hip_fp8_e4m3((float)xr[d - 1] * r).__x;
+  for (int k = (d8 << 3) + lane; k < d; k += 64)
+    qr[k] = __hip_fp8_e4m3((float)xr[k] * r).__x;
   if (lane == 0) {
     e8[row] = (unsigned char)(127 + k2);
     atomicMax(emax, 127 + k2);
