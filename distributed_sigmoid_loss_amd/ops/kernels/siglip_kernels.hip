@@ -925,9 +925,14 @@ __launch_bounds__(512) __global__ void quant_fp8_rowwise_kernel(
     const __bf16* __restrict__ x, unsigned char* __restrict__ q,
     unsigned char* __restrict__ e8, int* __restrict__ emax,
     int b, int d) {
-  const int row = blockIdx.x * 8 + (threadIdx.x >> 6);
-  if (row >= b) return;
+  // Grid-strided rows: one row per wave per iteration; a bounded grid
+  // keeps each wave looping (amortizes launch/drain and pipelines the
+  // per-row dependent load→reduce→convert chain across iterations).
   const int lane = threadIdx.x & 63;
+  const int rstep = gridDim.x * 8;
+  int emax_loc = 0;
+  for (int row = blockIdx.x * 8 + (threadIdx.x >> 6); row < b;
+       row += rstep) {
   const __bf16* xr = x + (size_t)row * d;
   const bf16x8* xr8 = reinterpret_cast<const bf16x8*>(xr);
   // vector path needs 16-B row alignment: d % 8 (rows then stay aligned)
@@ -965,14 +970,17 @@ __launch_bounds__(512) __global__ void quant_fp8_rowwise_kernel(
     qr[k] = __hip_fp8_e4m3((float)xr[k] * r).__x;
   if (lane == 0) {
     e8[row] = (unsigned char)(127 + k2);
-    atomicMax(emax, 127 + k2);
+    if (127 + k2 > emax_loc) emax_loc = 127 + k2;
   }
+  }
+  if (lane == 0 && emax_loc > 0) atomicMax(emax, emax_loc);
 }
 
 int quant_fp8_rowwise_bf16(uintptr_t stream, const void* x, void* q,
                            void* e8, void* emax, int b, int d) {
   if (b <= 0 || d <= 0) return (int)hipErrorInvalidValue;
-  hipLaunchKernelGGL(quant_fp8_rowwise_kernel, dim3(ceil_div(b, 8)),
+  const int blocks = ceil_div(b, 8) < 1024 ? ceil_div(b, 8) : 1024;
+  hipLaunchKernelGGL(quant_fp8_rowwise_kernel, dim3(blocks),
                      dim3(512), 0, (hipStream_t)stream, (const __bf16*)x,
                      (unsigned char*)q, (unsigned char*)e8, (int*)emax,
                      b, d);
