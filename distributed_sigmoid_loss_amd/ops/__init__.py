@@ -257,11 +257,11 @@ def rowwise_ok(b: int, n: int, d: int) -> bool:
     """Row-wise fp8 policy gate: needs the saved-g mm8-aligned shapes (the
     slab folds assume the fp8 GEMM path) and SIGLIP_FP8_ROWWISE=1.
 
-    OFF by default: it costs ~19%% of fp8 step throughput (4.94 vs
-    4.01 ms at B=32k — epilogue ratio folds + scale-reg path) and the
-    loss contract assumes unit-norm rows, where per-tensor scales lose
-    nothing.  Turn it on for embeddings with per-row dynamic range beyond
-    e4m3's ~2^18 (see test_rowwise_quant_keeps_per_row_precision)."""
+    OFF by default: it costs ~14%% of fp8 step throughput (4.84 vs
+    4.24 ms at B=32k — epilogue ratio folds + the row-wise quant pass)
+    and the loss contract assumes unit-norm rows, where per-tensor scales
+    lose nothing.  Turn it on for embeddings with per-row dynamic range
+    beyond e4m3's ~2^18 (see test_rowwise_quant_keeps_per_row_precision)."""
     return (os.environ.get("SIGLIP_FP8_ROWWISE", "0") == "1"
             and b % 16 == 0 and n % 16 == 0 and d % 16 == 0
             and hasattr(torch, "_scaled_mm"))
