@@ -62,8 +62,11 @@ class TwoTowerModel(nn.Module):
         self.text = ProjectionTower(in_dim, emb_dim)
 
     def forward(self, image_feats: torch.Tensor, text_feats: torch.Tensor):
+        import os
         from ..ops import l2_normalize
-        if (image_feats.is_cuda and image_feats.dtype == torch.bfloat16
+        if (os.environ.get("SIGLIP_BMM_TOWERS", "1") != "0"
+                and image_feats.is_cuda
+                and image_feats.dtype == torch.bfloat16
                 and image_feats.shape == text_feats.shape
                 and self.image.proj.weight.shape
                 == self.text.proj.weight.shape):
