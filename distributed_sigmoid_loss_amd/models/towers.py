@@ -49,11 +49,10 @@ class ProjectionTower(nn.Module):
 class TwoTowerModel(nn.Module):
     """Image + text projection towers sharing no weights.
 
-    On GPU with same-shaped bf16 inputs the two skinny projections run as
-    ONE strided-batched GEMM (the separate (32768, 768, 768) GEMMs measure
-    ~180 TF each on rocBLAS; batching roughly halves their wall time) —
-    autograd routes the batched grads back to each tower's weight through
-    the stack.
+    (Round 2 negative result, banked: batching the two skinny projections
+    into one strided-batched GEMM via stacked weights measured +0.15 ms —
+    the input-stack copies outweigh the batched-GEMM win at this shape.
+    Two plain GEMMs + the fused normalize kernels it is.)
     """
 
     def __init__(self, in_dim: int, emb_dim: int):
@@ -62,19 +61,4 @@ class TwoTowerModel(nn.Module):
         self.text = ProjectionTower(in_dim, emb_dim)
 
     def forward(self, image_feats: torch.Tensor, text_feats: torch.Tensor):
-        import os
-        from ..ops import l2_normalize
-        if (os.environ.get("SIGLIP_BMM_TOWERS", "1") != "0"
-                and image_feats.is_cuda
-                and image_feats.dtype == torch.bfloat16
-                and image_feats.shape == text_feats.shape
-                and self.image.proj.weight.shape
-                == self.text.proj.weight.shape):
-            w = torch.stack([self.image.proj.weight.mT,
-                             self.text.proj.weight.mT])       # (2, in, emb)
-            x = torch.stack([image_feats, text_feats])        # (2, b, in)
-            z = torch.bmm(x, w)                               # (2, b, emb)
-            b = image_feats.shape[0]
-            y = l2_normalize(z.view(2 * b, -1))   # one fused-normalize pass
-            return y[:b], y[b:]
         return self.image(image_feats), self.text(text_feats)
