@@ -66,18 +66,22 @@ def _world_and_rank(group=None):
 class _RingAllGatherLoss(torch.autograd.Function):
     """Fused distributed loss: ring-pipelined forward, reduce-scatter backward.
 
-    Forward walks the text shards around a unidirectional P2P ring; the
-    exchange of chunk k+1 is posted *before* the loss kernel for chunk k runs,
-    so on RCCL the xGMI wire time hides under the MFMA compute (the reference
-    serializes these — ``distributed_utils.py:25-26``).  All received shards
-    are kept (O(W·b·d), trivial against 288 GB HBM3E) so backward never
-    re-communicates embeddings.
+    Forward walks the text shards around a P2P ring — unidirectional, or
+    bidirectional with two xGMI links per hop and ⌈(W−1)/2⌉ hops; the
+    exchange of the next chunk is posted *before* the loss kernel for the
+    current one runs, so on RCCL the wire time hides under the MFMA compute
+    (the reference serializes these — ``distributed_utils.py:25-26``).  All
+    received shards are kept (O(W·b·d), trivial against 288 GB HBM3E) so
+    backward never re-communicates embeddings.  Each chunk's fwd+g kernel
+    writes its dL/dlogit block into one (b, W·b) slab (saved-g, the default)
+    so backward is pure GEMMs; the recompute path remains for huge batches.
 
-    Backward computes, per source shard s, the local gradient contribution
-    dztxt_s (recomputing logit tiles via the fused HIP kernel), concatenates
-    them in rank order and issues ONE ``reduce_scatter_tensor(SUM)`` — the
-    algebraic collapse of the reference's W−1 reversed autograd ring hops
-    (``distributed_utils.py:74-77``) into a single RCCL collective.
+    Backward produces the full text-gradient block and issues ONE
+    ``reduce_scatter_tensor(SUM)`` — the algebraic collapse of the
+    reference's W−1 reversed autograd ring hops
+    (``distributed_utils.py:74-77``) into a single RCCL collective —
+    launched async from the ``on_dztxt`` hook so its wire time overlaps the
+    remaining image-gradient GEMM.
 
     Gradient semantics match the differentiable all-gather strategy exactly:
     raw per-rank grads already carry full cross-rank contributions (what the
