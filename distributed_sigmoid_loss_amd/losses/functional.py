@@ -159,8 +159,13 @@ class _FusedSigmoidLoss(torch.autograd.Function):
       two GEMMs.  The logits GEMM runs once per step, like the reference's
       autograd (which saves the logits graph) but at O(b·n) g-slab memory
       instead of O(b·n) fp32 logits + labels.
-    - **recompute** (huge-batch chunked path, or SIGLIP_SAVE_G=0): forward
-      emits only the scalar; backward recomputes logit tiles slab by slab.
+    - **banded saved-g** (huge batches where ONE slab exceeds the kernel's
+      32-bit addressing but the full g fits HBM — 32 GB at 131k² vs
+      288 GB): forward emits g in column bands; backward is per-band GEMMs,
+      still no recompute.
+    - **recompute** (g would not fit HBM, explicit col_chunk, or
+      SIGLIP_SAVE_G=0): forward emits only the scalar; backward recomputes
+      logit tiles slab by slab.
     """
 
     @staticmethod
@@ -175,6 +180,9 @@ class _FusedSigmoidLoss(torch.autograd.Function):
         # Function.forward, so it cannot be probed here).
         save_g = (want_grad and col_chunk is None and zimg.is_cuda
                   and ops.save_g_enabled(b, n, quant))
+        banded = (not save_g and want_grad and col_chunk is None
+                  and zimg.is_cuda and ops.extension_available()
+                  and ops.save_g_banded_enabled(b, n, quant))
         qc = None
         if quant in ("fp8", "mixed") and zimg.is_cuda:
             # fp8 + saved-g on aligned shapes: ROW-wise pow2 scales, hardware
@@ -194,6 +202,24 @@ class _FusedSigmoidLoss(torch.autograd.Function):
             saved = (zimg, ztxt, t_prime, bias, out3, g) \
                 + ((gt,) if gt is not None else ()) \
                 + (qc if qc is not None else ())
+        elif banded:
+            step = ops.banded_col_step(b)
+            buf = ops._out_buf(zimg.device)
+            slabs = []
+            for j0 in range(0, n, step):
+                j1 = min(j0 + step, n)
+                g_s = torch.empty((b, j1 - j0), device=zimg.device,
+                                  dtype=torch.bfloat16)
+                diag = (None if diag_offset is None
+                        else int(diag_offset) - j0)
+                ops.siglip_fwd_g(zimg, ztxt[j0:j1].contiguous(), t_prime,
+                                 bias, diag, quant=quant, g_slab=g_s,
+                                 out3=buf)
+                slabs.append(g_s)
+            out3 = ops.reduce_out3(buf)
+            loss = out3[0].clone()
+            ctx.band_step = step
+            saved = (zimg, ztxt, t_prime, bias, out3) + tuple(slabs)
         else:
             loss = ops.siglip_fwd(zimg, ztxt, t_prime, bias, diag_offset,
                                   quant=quant, qcache=qc)
@@ -201,6 +227,7 @@ class _FusedSigmoidLoss(torch.autograd.Function):
                 + (qc if qc is not None else ())
         ctx.save_for_backward(*saved)
         ctx.saved_g = save_g
+        ctx.banded = banded
         ctx.has_gt = save_g and quant in ("fp8", "mixed")
         ctx.diag_offset = diag_offset
         ctx.col_chunk = col_chunk   # None → single slab when addressable
@@ -211,7 +238,28 @@ class _FusedSigmoidLoss(torch.autograd.Function):
     def backward(ctx, grad_output):
         from .. import ops
         zimg, ztxt, t_prime, bias = ctx.saved_tensors[:4]
-        if ctx.saved_g:
+        if getattr(ctx, "banded", False):
+            out3 = ctx.saved_tensors[4]
+            slabs = ctx.saved_tensors[5:]
+            step = ctx.band_step
+            dev = zimg.device
+            tp32 = t_prime.detach().reshape(()).to(dev).float()
+            t_true = tp32.exp()
+            go = grad_output.detach().reshape(()).to(dev).float()
+            scale = go * t_true
+            dzimg_acc = torch.zeros_like(zimg, dtype=torch.float32)
+            dztxt = torch.empty_like(ztxt)
+            for k, g_s in enumerate(slabs):
+                j0 = k * step
+                j1 = min(j0 + step, ztxt.shape[0])
+                zt_s = ztxt[j0:j1]
+                dzimg_acc += ((g_s @ zt_s) * scale).float()
+                dztxt[j0:j1] = ((g_s.T @ zimg) * scale).to(ztxt.dtype)
+            dzimg = dzimg_acc.to(zimg.dtype)
+            dt_prime = (out3[1] * go * t_true).to(
+                t_prime.dtype).reshape(t_prime.shape)
+            dbias = (out3[2] * go).to(bias.dtype).reshape(bias.shape)
+        elif ctx.saved_g:
             i = 4
             out3, g = ctx.saved_tensors[i:i + 2]
             i += 2

@@ -537,6 +537,39 @@ def save_g_enabled(b: int, n: int, quant: str) -> bool:
     return need <= cap
 
 
+def save_g_banded_enabled(b: int, n: int, quant: str) -> bool:
+    """Column-banded saved-g policy: when ONE slab exceeds the kernel's
+    32-bit addressing but the full (b, n) g fits HBM comfortably (it is
+    32 GB at 131k², against 288 GB), forward emits g in column bands and
+    backward is still pure GEMMs — no logits recompute.  bf16 only (the
+    fp8 slabs come in pairs and ride the single-slab path);
+    SIGLIP_SAVE_G_BANDED=0 disables, SIGLIP_SAVE_G=0 disables all saving."""
+    if quant != "bf16":
+        return False
+    if os.environ.get("SIGLIP_SAVE_G", "auto") == "0":
+        return False
+    if os.environ.get("SIGLIP_SAVE_G_BANDED", "auto") == "0":
+        return False
+    if not torch.cuda.is_available():
+        return False
+    total = b * n * 2
+    try:
+        free, _ = torch.cuda.mem_get_info()
+    except Exception:  # pragma: no cover
+        return False
+    return total <= free * 0.5
+
+
+def banded_col_step(b: int, esz: int = 2) -> int:
+    """Widest 256-aligned column band whose slab stays under the kernel's
+    32-bit store addressing (SIGLIP_BANDED_STEP overrides, for tests)."""
+    env = os.environ.get("SIGLIP_BANDED_STEP")
+    if env:
+        return max(256, int(env))
+    step = (2 ** 32 - 1) // (b * esz)
+    return max(256, (step // 256) * 256)
+
+
 def scaled_mm8(a8: torch.Tensor, b8_rowmajor: torch.Tensor,
                scale: torch.Tensor) -> torch.Tensor:
     """fp8 GEMM a8 (m, k) @ b8 (k, d) → bf16 via hipBLASLt

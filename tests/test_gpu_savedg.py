@@ -258,6 +258,34 @@ def test_fused_quant_matches_torch(b, d):
     assert (diff / denom).max().item() < 0.07, (diff / denom).max()
 
 
+def test_banded_savedg_matches_recompute():
+    """Column-banded saved-g (the huge-batch regime, forced small here via
+    env) == recompute backward == fp32 reference."""
+    b, n, d = 512, 1024, 768
+    outs = {}
+    for mode, env in (("banded", {"SIGLIP_SAVE_G_MAX_BYTES": "1000",
+                                  "SIGLIP_BANDED_STEP": "256"}),
+                      ("recompute", {"SIGLIP_SAVE_G": "0"})):
+        for k, v in env.items():
+            os.environ[k] = v
+        try:
+            zi, zt, tp, bs = make_inputs(b, n, d, seed=77)
+            zi = zi.clone().requires_grad_(True)
+            zt = zt.clone().requires_grad_(True)
+            tp = tp.clone().requires_grad_(True)
+            bs = bs.clone().requires_grad_(True)
+            loss = sigmoid_contrastive_loss(zi, zt, tp, bs, diag_offset=256)
+            loss.backward()
+            torch.cuda.synchronize()
+            outs[mode] = (loss.detach(), zi.grad, zt.grad, tp.grad, bs.grad)
+        finally:
+            for k in env:
+                os.environ.pop(k, None)
+    for a, b_ in zip(outs["banded"], outs["recompute"]):
+        assert torch.allclose(a.float(), b_.float(), rtol=2e-2, atol=1e-3), \
+            (a.float() - b_.float()).abs().max()
+
+
 def test_inference_skips_slab():
     """Under no_grad the plain forward kernel runs (no g allocation) — the
     loss is identical either way."""
