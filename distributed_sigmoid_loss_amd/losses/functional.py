@@ -21,9 +21,12 @@ Backward (wrt zimg, ztxt, t_prime, bias), with g_ij = −l_ij·σ(−l_ij·z_ij)
 
 On a GPU, dispatch goes to the hand-written CDNA4 kernels in
 ``distributed_sigmoid_loss_amd.ops`` (MFMA logits, fused epilogue, fp32
-accumulation; backward recomputes logit tiles instead of saving them).  On CPU
-(and only on CPU) a plain differentiable PyTorch path is used — on a GPU a
-missing extension raises rather than silently falling back.
+accumulation).  By default the forward also emits the dL/dlogit slab
+("saved-g" — single-slab or column-banded) so backward is pure GEMMs; the
+recompute-backward kernels serve explicit ``col_chunk`` runs and batches
+whose g would not fit HBM.  On CPU (and only on CPU) a plain differentiable
+PyTorch path is used — on a GPU a missing extension raises rather than
+silently falling back.
 
 Column chunking: ``col_chunk`` bounds the working set to ``O(b·col_chunk)`` so
 per-GPU batches of 131072+ (BASELINE config 4) never materialize a full

@@ -5,15 +5,20 @@ exposes:
 
 - :func:`siglip_fwd` — fused logits+logsigmoid+sum forward over a (b, n)
   block; the logits matrix never leaves MFMA accumulators.
-- :func:`siglip_bwd` — backward: fused recompute kernel emits the
-  dL/dlogit slab(s) and the scalar partials; the two ``(b,n)×(n,d)``
-  gradient GEMMs run on the GEMM libraries (rocBLAS bf16 via
-  ``torch.matmul``, or hipBLASLt fp8 via ``torch._scaled_mm`` for the
-  fp8/mixed policies, consuming the kernel-emitted e4m3 g and gᵀ slabs),
-  column-chunked so workspace is O(b · col_chunk) regardless of n.
+- :func:`siglip_fwd_g` / :func:`siglip_bwd_from_g` — the default training
+  pair ("saved-g"): one kernel emits loss + the dL/dlogit slab + both
+  scalar partials, and backward is just the two ``(b,n)×(n,d)`` gradient
+  GEMMs on the libraries (rocBLAS bf16 via ``torch.matmul``, tuned
+  hipBLASLt fp8 via ``torch._scaled_mm``).
+- :func:`siglip_bwd` — recompute backward for explicit ``col_chunk`` runs
+  and batches whose g slab would not fit: the recompute kernel re-derives
+  logit tiles and emits g slab by slab, O(b · col_chunk) workspace.
 - quantization policies: ``bf16`` | ``fp8`` (e4m3 logits via the MX-scaled
-  MFMA, per-tensor scales folded into the temperature) | ``mixed`` (bf16
+  MFMA — per-tensor scales folded into the temperature, or per-row e8m0
+  hardware dequant with ``SIGLIP_FP8_ROWWISE=1``) | ``mixed`` (bf16
   logits, fp8 gradient GEMMs).
+- fused tower helpers: :func:`l2_normalize` (single-pass fwd/bwd) and the
+  fused fp8 quantizers.
 
 These are *loud* paths: calling them on a GPU without the built extension
 raises — there is no silent eager fallback on device (CPU fallbacks live in
