@@ -169,6 +169,19 @@ def test_rccl_fp8_ring_w2():
 
 
 @needs2
+def test_rccl_mixed_ring_w2():
+    """mixed policy (bf16 logits, fp8 grad GEMMs) on the RCCL ring at W=2
+    vs the bf16 ring, fp8-grad-class tolerance."""
+    mixed = run_rccl(2, "ring", quant="mixed", average=False)
+    bf16 = run_rccl(2, "ring", average=False)
+    assert torch.allclose(mixed["loss"].float(), bf16["loss"].float(),
+                          rtol=2e-2)
+    for key in ("zi", "zt"):
+        assert torch.allclose(mixed[key].float(), bf16[key].float(),
+                              rtol=2e-1, atol=2e-2), key
+
+
+@needs2
 def test_rccl_savedg_vs_recompute_w2():
     """saved-g ring backward == recompute ring backward at W=2 on RCCL."""
     os.environ["SIGLIP_SAVE_G"] = "1"
