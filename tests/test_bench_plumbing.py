@@ -65,3 +65,14 @@ def test_bench_distributed_cpu_2proc_allgather_ddp():
          "--ddp"],
         capture_output=True, text=True, timeout=600, cwd=REPO)
     assert res.returncode == 0, res.stderr[-2000:]
+
+
+def test_example_cpu_smoke():
+    """examples/train_siglip.py runs end to end on CPU (3 steps, tiny)."""
+    res = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples", "train_siglip.py"),
+         "--device", "cpu", "--batch-per-gpu", "8", "--dim", "32",
+         "--steps", "3", "--log-every", "2"],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert res.returncode == 0, res.stderr[-2000:]
+    assert "done:" in res.stdout
