@@ -35,7 +35,6 @@ per-GPU batches of 131072+ (BASELINE config 4) never materialize a full
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

@@ -20,7 +20,7 @@ double-buffered schedule (the reference serializes comm and compute by calling
 from __future__ import annotations
 
 import os
-from typing import Optional, Sequence
+from typing import Sequence
 
 import torch
 import torch.distributed as dist
