@@ -251,12 +251,13 @@ class _FusedSigmoidLoss(torch.autograd.Function):
             scale = go * t_true
             dzimg_acc = torch.zeros_like(zimg, dtype=torch.float32)
             dztxt = torch.empty_like(ztxt)
+            scale_c = scale.to(ztxt.dtype)
             for k, g_s in enumerate(slabs):
                 j0 = k * step
                 j1 = min(j0 + step, ztxt.shape[0])
                 zt_s = ztxt[j0:j1]
                 dzimg_acc += ((g_s @ zt_s) * scale).float()
-                dztxt[j0:j1] = ((g_s.T @ zimg) * scale).to(ztxt.dtype)
+                dztxt[j0:j1] = (g_s.T @ zimg) * scale_c
             dzimg = dzimg_acc.to(zimg.dtype)
             dt_prime = (out3[1] * go * t_true).to(
                 t_prime.dtype).reshape(t_prime.shape)

@@ -763,10 +763,15 @@ def siglip_bwd_from_g(zimg: torch.Tensor, ztxt: torch.Tensor,
             dzimg = (g16 @ ztxt) * scale
         t_eff = t_true * si * st if quant == "fp8" else t_true
     else:
-        dztxt = (g.T @ zimg) * scale
+        # Scale in the GEMM output's own dtype: a fp32 0-d multiplier
+        # promotes the whole (n, d) product to fp32 (two extra full-tensor
+        # passes for the round trip); the bf16 scalar rounding (~0.4%) is
+        # below the bf16 grad noise floor.
+        scale_c = scale.to(g.dtype)
+        dztxt = (g.T @ zimg) * scale_c
         if on_dztxt is not None:
             on_dztxt(dztxt)
-        dzimg = (g @ ztxt) * scale
+        dzimg = (g @ ztxt) * scale_c
         t_eff = t_true
     dt_prime = (out3[1] * go * t_eff).to(t_prime.dtype).reshape(t_prime.shape)
     dbias = (out3[2] * go).to(bias.dtype).reshape(bias.shape)
