@@ -17,13 +17,14 @@ opt = torch.optim.SGD(list(model.parameters()) + list(loss_mod.parameters()),
 img = torch.randn(b, d, device="cuda", dtype=torch.bfloat16)
 txt = torch.randn(b, d, device="cuda", dtype=torch.bfloat16)
 t0 = time.perf_counter()
-for step in range(300):
+N = int(sys.argv[2]) if len(sys.argv) > 2 else 300
+for step in range(N):
     opt.zero_grad(set_to_none=True)
     zi, zt = model(img, txt)
     loss = loss_mod(zi, zt)
     loss.backward()
     opt.step()
-    if (step + 1) % 50 == 0:
+    if (step + 1) % 100 == 0 or step == 49:
         torch.cuda.synchronize()
         v = float(loss.detach())
         assert v == v and v > 0, f"bad loss {v} at step {step}"
@@ -32,4 +33,4 @@ for step in range(300):
               f"peak={torch.cuda.max_memory_allocated()/2**30:.2f}GiB",
               flush=True)
 el = time.perf_counter() - t0
-print(f"300 steps OK ({quant}), {el:.1f}s wall, {b*300/el/1e6:.2f}M pairs/s")
+print(f"{N} steps OK ({quant}), {el:.1f}s wall, {b*N/el/1e6:.2f}M pairs/s")
