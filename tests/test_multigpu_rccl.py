@@ -113,6 +113,19 @@ def run_rccl(world, strategy, quant="bf16", bidir=False, average=True):
 
 needs2 = pytest.mark.skipif(N_GPUS < 2, reason="needs >=2 GPUs")
 needs4 = pytest.mark.skipif(N_GPUS < 4, reason="needs >=4 GPUs")
+needs8 = pytest.mark.skipif(N_GPUS < 8, reason="needs 8 GPUs")
+
+
+@needs8
+@pytest.mark.parametrize("bidir", [False, True])
+def test_rccl_full_node_w8(bidir):
+    """Full 8-GPU node: ring (both directions — bidir runs 3 rounds + a
+    remainder hop) vs all-gather, raw grads."""
+    ring = run_rccl(8, "ring", bidir=bidir, average=False)
+    gather = run_rccl(8, "all_gather", average=False)
+    for key in ("zi", "zt", "t_prime", "bias", "loss"):
+        assert torch.allclose(ring[key].float(), gather[key].float(),
+                              rtol=2e-2, atol=5e-4), key
 
 
 @needs2

@@ -82,6 +82,16 @@ def test_bidir_ring_matches_all_gather_raw_grads(world):
         assert torch.allclose(ring[key], gather[key], rtol=1e-4, atol=1e-7), key
 
 
+def test_bidir_ring_w5_rotation():
+    """W=5: TWO bidirectional rounds, exercising the re-post of received
+    chunks (the rotation ``to_left, to_right ← received pair``) that W≤4
+    never reaches."""
+    ring = run_distributed(ddp_step, 5, 2, 16, "ring_bidir", False)[0]
+    gather = run_distributed(ddp_step, 5, 2, 16, "all_gather", False)[0]
+    for key in ("img", "txt", "t_prime", "bias", "loss"):
+        assert torch.allclose(ring[key], gather[key], rtol=1e-4, atol=1e-7), key
+
+
 @pytest.mark.parametrize("world,batch", [(3, 3), (4, 2)])
 def test_bidir_ring_scaling_oracle(world, batch):
     """N-rank bidir-ring grads == 1-rank grads (the reference's main
