@@ -42,6 +42,8 @@ from ..parallel.ring import (
     neighbour_exchange_bidir_with_grad,
     neighbour_exchange_start,
     neighbour_exchange_bidir_start,
+    quantized_exchange_start,
+    quantized_exchange_bidir_start,
 )
 from ..utils.profiling import roctx_range, HopStats
 
@@ -112,22 +114,37 @@ class _RingAllGatherLoss(torch.autograd.Function):
                   and _ops.extension_available()
                   and _ops.save_g_enabled(b_img, world * b_txt, quant))
 
-        # fp8: quantize the image shard ONCE for all W chunk losses; each
-        # received text chunk gets its own per-tensor scale, SAVED for
-        # backward so fwd and bwd see identical quantized values (the
+        # fp8: quantize BOTH own shards ONCE and ship e4m3 + scale over the
+        # wire — half the xGMI bytes per hop versus bf16, no per-chunk
+        # quantize kernels, and every rank computes with the exact
+        # quantized values the owner produced.  Per-chunk scales are SAVED
+        # for backward so fwd and bwd see identical quantized values (the
         # round-1 version requantized the concatenated block with one
         # scale — fwd/bwd loss surfaces differed at W>1).
+        fp8_wire = quant == "fp8" and world > 1
         use_q = quant == "fp8" and on_gpu
         qcaches = [None] * world        # per-source (zi_q, si, zt_q, st)
-        if use_q:
+        if quant == "fp8":
             zi_q, si = _ops._quant_fp8(zimg)
+            zt_q0, st0 = _ops._quant_fp8(ztxt)
 
-            def qc_for(zt_chunk):
-                zt_q, st = _ops._quant_fp8(zt_chunk)
-                return (zi_q, si, zt_q, st)
+            def dequant(q8, sc):
+                return (q8.to(torch.float32) * sc).to(zimg.dtype)
+
+            def qc_of(q8, sc):
+                return (zi_q, si, q8, sc.reshape(()))
         else:
-            def qc_for(zt_chunk):
+            zt_q0 = st0 = None
+
+        def qc_for(zt_chunk):
+            # own-rank / non-wire chunks only (wire chunks carry their
+            # quantization with them)
+            if quant != "fp8":
                 return None
+            if zt_chunk is ztxt:
+                return qc_of(zt_q0, st0)
+            zt_q, st = _ops._quant_fp8(zt_chunk)
+            return (zi_q, si, zt_q, st)
 
         g_slab = gt_slab = out3 = None
         g_chunks = [None] * world       # fp8: per-chunk g (scales differ)
@@ -143,8 +160,9 @@ class _RingAllGatherLoss(torch.autograd.Function):
                 gt_slab = torch.empty((world * b_txt, b_img),
                                       device=zimg.device, dtype=g_dtype)
 
-        def chunk_fwd(zt_chunk, src, diag):
-            qc = qc_for(zt_chunk)
+        def chunk_fwd(zt_chunk, src, diag, qc=None):
+            if qc is None:
+                qc = qc_for(zt_chunk)
             qcaches[src] = qc
             if not save_g:
                 return chunk_loss_fwd(zimg, zt_chunk, t_prime, bias,
@@ -162,6 +180,38 @@ class _RingAllGatherLoss(torch.autograd.Function):
                 g_slab=g_slab, gt_slab=gt_slab, col0=src * b_txt, out3=out3)
             return None   # shared out3 accumulates across chunks
 
+        # Wire payload abstraction: bf16/mixed ship the chunk tensor; fp8
+        # ships its (e4m3-as-uint8, fp32 scale) pair.  unpack() returns the
+        # compute-dtype chunk plus the per-chunk qcache.
+        if fp8_wire:
+            def wire_uni(payload):
+                return quantized_exchange_start(left, right, payload[0],
+                                                payload[1], group=group)
+
+            def wire_bidir(pl, pr):
+                return quantized_exchange_bidir_start(
+                    left, right, pl[0], pl[1], pr[0], pr[1], group=group)
+
+            def unpack(recvs):
+                q8 = recvs[0].view(torch.float8_e4m3fn)
+                sc = recvs[1]
+                return dequant(q8, sc), qc_of(q8, sc), (q8, sc)
+
+            own_payload = (zt_q0, st0.reshape(1).float())
+        else:
+            def wire_uni(payload):
+                return neighbour_exchange_start(left, right, payload,
+                                                group=group)
+
+            def wire_bidir(pl, pr):
+                return neighbour_exchange_bidir_start(left, right, pl, pr,
+                                                      group=group)
+
+            def unpack(recvs):
+                return recvs[0], None, recvs[0]
+
+            own_payload = ztxt
+
         use_bidir = bidir and world > 2
         if world > 1:
             left = (rank - 1 + world) % world
@@ -171,14 +221,11 @@ class _RingAllGatherLoss(torch.autograd.Function):
             # TWO xGMI point-to-point links per hop (reference
             # distributed_utils.py:30-62 at the perf layer), halving the hop
             # count to ⌈(W−1)/2⌉ — at N=8 per-rank b=4096 the ring is
-            # wire-bound (~42 µs/hop vs ~40 µs chunk compute), so hop count
-            # is the lever.
+            # wire-bound, so hop count and bytes per hop are the levers.
             if use_bidir:
-                handle = neighbour_exchange_bidir_start(left, right, ztxt,
-                                                        ztxt, group=group)
+                handle = wire_bidir(own_payload, own_payload)
             else:
-                handle = neighbour_exchange_start(left, right, ztxt,
-                                                  group=group)
+                handle = wire_uni(own_payload)
 
         loss = chunk_fwd(ztxt, rank, 0)
 
@@ -187,46 +234,49 @@ class _RingAllGatherLoss(torch.autograd.Function):
             if part is not None:
                 loss = loss + part
 
+        def take(recvs, src):
+            chunk, qc, payload = unpack(recvs)
+            chunks[src] = chunk
+            return chunk, qc, payload
+
         if world > 1 and use_bidir:
             nb, rem = divmod(world - 1, 2)
+            npl = 2 if fp8_wire else 1    # tensors per direction on the wire
             for r in range(1, nb + 1):
                 with _hop_span(f"ring_bhop{r}_wait"):
-                    from_right, from_left = handle.wait()
+                    recvs = handle.wait()
                 src_r = (rank + r) % world
                 src_l = (rank - r + world) % world
-                chunks[src_r] = from_right
-                chunks[src_l] = from_left
+                from_right, qc_r, pl_r = take(recvs[:npl], src_r)
+                from_left, qc_l, pl_l = take(recvs[npl:], src_l)
                 if r < nb:
-                    handle = neighbour_exchange_bidir_start(
-                        left, right, from_right, from_left, group=group)
+                    handle = wire_bidir(pl_r, pl_l)
                 elif rem:
                     # one unidirectional remainder hop: forward the
                     # rightward-traveling stream (last received from left)
-                    handle = neighbour_exchange_start(left, right, from_left,
-                                                      group=group)
+                    handle = wire_uni(pl_l)
                 with _hop_span(f"ring_bchunk{r}_loss"):
-                    add_part(chunk_fwd(from_right, src_r, None))
-                    add_part(chunk_fwd(from_left, src_l, None))
+                    add_part(chunk_fwd(from_right, src_r, None, qc_r))
+                    add_part(chunk_fwd(from_left, src_l, None, qc_l))
             if rem:
                 with _hop_span("ring_rem_wait"):
-                    recv = handle.wait()[0]
+                    recvs = handle.wait()
                 src = (rank - nb - 1 + world) % world
-                chunks[src] = recv
-                add_part(chunk_fwd(recv, src, None))
+                recv, qc, _ = take(recvs, src)
+                add_part(chunk_fwd(recv, src, None, qc))
         elif world > 1:
             for hop in range(1, world):
                 # Post hop k+1 before computing on hop k's data.  roctx
                 # ranges label the hops for rocprofv3/torch.profiler traces
                 # (per-hop comm visibility).
                 with _hop_span(f"ring_hop{hop}_wait"):
-                    recv = handle.wait()[0]
+                    recvs = handle.wait()
                 src = (rank - hop + world) % world
-                chunks[src] = recv
+                recv, qc, payload = take(recvs, src)
                 if hop < world - 1:
-                    handle = neighbour_exchange_start(left, right, recv,
-                                                      group=group)
+                    handle = wire_uni(payload)
                 with _hop_span(f"ring_chunk{hop}_loss"):
-                    add_part(chunk_fwd(recv, src, None))
+                    add_part(chunk_fwd(recv, src, None, qc))
 
         if save_g:
             # Reduce the per-XCD scalar buffers once, after every chunk's

@@ -15,12 +15,14 @@ from distributed_sigmoid_loss_amd.parallel import average_gradients
 from helpers import encode_shard, run_distributed
 
 
-def ddp_step(rank, world, batch_per_rank, emb_dim, strategy, average):
+def ddp_step(rank, world, batch_per_rank, emb_dim, strategy, average,
+             quant="bf16"):
     img_enc, txt_enc, zi, zt = encode_shard(rank, world, batch_per_rank,
                                             emb_dim)
     bidir = strategy == "ring_bidir"
     loss_mod = DistributedSigmoidLoss(
-        batch_per_rank, strategy="ring" if bidir else strategy, bidir=bidir)
+        batch_per_rank, strategy="ring" if bidir else strategy, bidir=bidir,
+        quant=quant)
     loss = loss_mod(zi, zt)
     loss.backward()
     if average:
@@ -101,3 +103,19 @@ def test_bidir_ring_scaling_oracle(world, batch):
                              True)[0]
     for key in ("img", "txt", "t_prime", "bias"):
         assert torch.allclose(multi[key], single[key], rtol=1e-3, atol=1e-6), key
+
+
+@pytest.mark.parametrize("world,strategy", [(3, "ring"), (4, "ring_bidir")])
+def test_fp8_wire_ring_cpu_routing(world, strategy):
+    """fp8 ring ships e4m3 + scale over the wire; on CPU the received
+    chunks are dequantized for the torch path.  Validates the wire
+    routing/scale pairing (incl. bidir forwarding) against the all_gather
+    strategy — which on CPU computes on unquantized values — so the
+    tolerance is the e4m3 quantization level, not exactness."""
+    ring = run_distributed(ddp_step, world, 4, 32, strategy, False, "fp8")[0]
+    gather = run_distributed(ddp_step, world, 4, 32, "all_gather", False,
+                             "bf16")[0]
+    for key in ("img", "txt", "loss"):
+        a, b = ring[key].float(), gather[key].float()
+        rel = (a - b).norm() / b.norm().clamp(min=1e-12)
+        assert rel < 8e-2, (key, rel)
