@@ -12,10 +12,9 @@
 // with labels l_ij = +1 iff j == i + diag_offset (an index predicate — the
 // (b,n) label matrix of the reference, distributed_sigmoid_loss.py:28-30 and
 // rwightman_sigmoid_loss.py:43-47, is never materialized), l_ij = -1 else.
-// The (b,n) logits matrix never leaves the MFMA accumulators in forward; the
-// backward recomputes it tile-by-tile and emits only the g slab consumed by
-// the two rocBLAS GEMMs (dzimg = t·g@ztxt, dztxt = t·gᵀ@zimg) on the Python
-// side (ops/__init__.py).
+// The (b,n) logits matrix never leaves the MFMA accumulators; only the g
+// slab consumed by the two library GEMMs (dzimg = t·g@ztxt,
+// dztxt = t·gᵀ@zimg, ops/__init__.py) ever reaches memory.
 //
 // MODE 2 ("fwd+g") fuses the two: one kernel emits the loss, the g slab and
 // both scalar partials, so a training step never computes the logits GEMM
