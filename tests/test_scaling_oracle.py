@@ -119,3 +119,13 @@ def test_fp8_wire_ring_cpu_routing(world, strategy):
         a, b = ring[key].float(), gather[key].float()
         rel = (a - b).norm() / b.norm().clamp(min=1e-12)
         assert rel < 8e-2, (key, rel)
+
+
+def test_bidir_ring_w8_scale_shape():
+    """W=8 — the exact control-flow shape of the driver's 8-GPU SCALE run:
+    THREE bidirectional rounds (two chunk re-posts) plus the unidirectional
+    remainder hop."""
+    ring = run_distributed(ddp_step, 8, 2, 16, "ring_bidir", False)[0]
+    gather = run_distributed(ddp_step, 8, 2, 16, "all_gather", False)[0]
+    for key in ("img", "txt", "t_prime", "bias", "loss"):
+        assert torch.allclose(ring[key], gather[key], rtol=1e-4, atol=1e-7), key
