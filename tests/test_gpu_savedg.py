@@ -258,6 +258,32 @@ def test_fused_quant_matches_torch(b, d):
     assert (diff / denom).max().item() < 0.07, (diff / denom).max()
 
 
+def test_fp8_wire_simulated_roundtrip_gpu():
+    """The fp8 ring's wire math on one GPU: quantize a chunk, 'transmit' its
+    uint8 view + scale, reconstruct the receiver's qcache, and compute —
+    must match computing directly from the sender's quantization."""
+    b, n, d = 512, 512, 768
+    zi, zt, tp, bs = make_inputs(b, n, d, seed=31)
+    zi_q, si = ops._quant_fp8(zi)
+    zt_q, st = ops._quant_fp8(zt)
+    # "wire": uint8 payload + (1,) fp32 scale, as parallel.ring ships them
+    wire_q = zt_q.view(torch.uint8).contiguous().clone()
+    wire_s = st.reshape(1).float().clone()
+    rq = wire_q.view(torch.float8_e4m3fn)
+    rs = wire_s.reshape(())
+    assert torch.equal(rq, zt_q) and torch.equal(rs, st.reshape(()))
+
+    direct = ops.siglip_fwd_g(zi, zt, tp, bs, None, quant="fp8",
+                              qcache=(zi_q, si, zt_q, st))
+    via_wire = ops.siglip_fwd_g(zi, (rq.to(torch.float32) * rs).to(zt.dtype),
+                                tp, bs, None, quant="fp8",
+                                qcache=(zi_q, si, rq, rs))
+    torch.cuda.synchronize()
+    assert torch.equal(ops.reduce_out3(direct[0]), ops.reduce_out3(via_wire[0]))
+    assert torch.equal(direct[1], via_wire[1])          # g slabs bitwise
+    assert torch.equal(direct[2], via_wire[2])          # gt slabs bitwise
+
+
 def test_banded_savedg_matches_recompute():
     """Column-banded saved-g (the huge-batch regime, forced small here via
     env) == recompute backward == fp32 reference."""
