@@ -181,6 +181,53 @@ def test_rccl_fp8_ring_w2():
                               rtol=2e-1, atol=2e-3), key
 
 
+def _siglip_worker(rank, world, port, bidir, ret):
+    import math
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.cuda.set_device(rank)
+    dist.init_process_group("nccl", rank=rank, world_size=world)
+    try:
+        from distributed_sigmoid_loss_amd import SigLipLoss
+        zi_all, zt_all = _global_batch(world)
+        zi = zi_all[rank * B:(rank + 1) * B].cuda(rank).requires_grad_(True)
+        zt = zt_all[rank * B:(rank + 1) * B].cuda(rank).requires_grad_(True)
+        sc = torch.tensor(math.log(10.0), device=f"cuda:{rank}",
+                          requires_grad=True)
+        bs = torch.tensor(-10.0, device=f"cuda:{rank}", requires_grad=True)
+        mod = SigLipLoss(rank=rank, world_size=world, bidir=bidir)
+        loss = mod(zi, zt, sc, bs)
+        loss.backward()
+        torch.cuda.synchronize()
+        if rank == 0:
+            ret["out"] = {"zi": zi.grad.cpu(), "zt": zt.grad.cpu(),
+                          "t_prime": sc.grad.cpu(), "bias": bs.grad.cpu(),
+                          "loss": loss.detach().cpu()}
+    finally:
+        dist.destroy_process_group()
+
+
+@needs2
+@pytest.mark.parametrize("bidir", [True, False])
+def test_rccl_sigliploss_compat_w2(bidir):
+    """The reference-API SigLipLoss (caller-owned scale/bias, autograd
+    exchange chain) on real RCCL at W=2 vs DistributedSigmoidLoss's
+    all-gather strategy: the strategy-equivalence oracle on hardware.
+    SigLipLoss normalizes per chunk by b; DistributedSigmoidLoss divides
+    the total by b — identical algebra, so raw grads must match."""
+    manager = mp.Manager()
+    ret = manager.dict()
+    mp.spawn(_siglip_worker, args=(2, free_port(), bidir, ret), nprocs=2,
+             join=True)
+    rw = ret["out"]
+    gather = run_rccl(2, "all_gather", average=False)
+    # gather path's loss is total/b; SigLipLoss sums per-chunk/b — equal.
+    for key in ("zi", "zt", "t_prime", "bias", "loss"):
+        assert torch.allclose(rw[key].float(), gather[key].float(),
+                              rtol=2e-2, atol=5e-4), key
+
+
 @needs2
 def test_rccl_mixed_ring_w2():
     """mixed policy (bf16 logits, fp8 grad GEMMs) on the RCCL ring at W=2
