@@ -5,6 +5,8 @@ from .ring import (
     neighbour_exchange_bidir_with_grad,
     neighbour_exchange_start,
     neighbour_exchange_bidir_start,
+    quantized_exchange_start,
+    quantized_exchange_bidir_start,
     NeighbourExchange,
     NeighbourExchangeBidir,
     RingHandle,
@@ -18,6 +20,8 @@ __all__ = [
     "neighbour_exchange_bidir_with_grad",
     "neighbour_exchange_start",
     "neighbour_exchange_bidir_start",
+    "quantized_exchange_start",
+    "quantized_exchange_bidir_start",
     "NeighbourExchange",
     "NeighbourExchangeBidir",
     "RingHandle",
