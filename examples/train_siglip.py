@@ -55,7 +55,7 @@ def main():
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--strategy", choices=["ring", "all_gather"],
                    default="ring")
-    p.add_argument("--quant", choices=["bf16", "fp8"], default="bf16")
+    p.add_argument("--quant", choices=["bf16", "fp8", "mixed"], default="bf16")
     p.add_argument("--device", default=None)
     p.add_argument("--log-every", type=int, default=10)
     args = p.parse_args()
