@@ -93,6 +93,10 @@ def load_tuned_gemms() -> bool:
     global _tuned_loaded
     if _tuned_loaded:
         return True
+    if os.environ.get("PYTORCH_TUNABLEOP_TUNING", "0") == "1":
+        # Explicit tuning session (e.g. regenerating the table): leave
+        # torch's env-driven TunableOp configuration alone.
+        return False
     if not torch.cuda.is_available():
         return False
     path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
