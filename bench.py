@@ -61,10 +61,13 @@ def parse_args():
                    help="capture the steady-state step in a hipGraph and "
                         "replay it (single-GPU; falls back to eager if "
                         "capture fails)")
-    p.add_argument("--ddp", action="store_true",
+    p.add_argument("--ddp", dest="ddp", action="store_true", default=True,
                    help="wrap the towers in torch DDP (bucketed RCCL "
-                        "all-reduce overlapped with backward) instead of "
-                        "manual grad averaging")
+                        "all-reduce overlapped with backward) — the "
+                        "default at N>1")
+    p.add_argument("--no-ddp", dest="ddp", action="store_false",
+                   help="manual post-backward grad averaging instead "
+                        "(the reference harness's scheme)")
     return p.parse_args()
 
 
