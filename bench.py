@@ -7,9 +7,10 @@ bf16, strong scaling over 1/2/4/8 MI355X GPUs (per-rank batch = B/N; the
 pair-grid work B² is fixed as N grows).
 
 One step = encode both towers → L2-normalize → DistributedSigmoidLoss
-(fused CDNA4 HIP kernels; ring or all-gather comm over RCCL/xGMI) → backward →
-manual DDP gradient averaging → SGD step.  Data is synthetic random features;
-weights are random-init (no datasets/checkpoints are available offline).
+(fused CDNA4 HIP kernels; ring or all-gather comm over RCCL/xGMI) → backward
+(tower grads via torch DDP's bucketed all-reduce overlapped with backward at
+N>1; loss params averaged manually) → SGD step.  Data is synthetic random
+features; weights are random-init (no datasets/checkpoints offline).
 
 Launch (driver contract):
     python bench.py --gpus N --steps K --warmup W
