@@ -20,7 +20,7 @@ The compute path is PyTorch-ROCm + HIP/CDNA4 kernels + RCCL over xGMI; there
 are no CUDA compatibility layers and no multi-backend dispatch.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from .losses.sigmoid_loss import DistributedSigmoidLoss, SigLipLoss
 from .losses.functional import sigmoid_contrastive_loss
