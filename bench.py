@@ -195,6 +195,7 @@ def main():
                 "parallelism": f"dp{world}-{args.strategy}",
                 "impl": args.impl,
                 "graph": bool(use_graph),
+                "ddp_towers": bool(args.ddp and world > 1),
             },
         }), flush=True)
 
