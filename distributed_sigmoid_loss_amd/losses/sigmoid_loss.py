@@ -216,6 +216,12 @@ class _RingAllGatherLoss(torch.autograd.Function):
         if world > 1:
             left = (rank - 1 + world) % world
             right = (rank + 1) % world
+            if group is not None:
+                # P2POp peers are GLOBAL ranks even when a group is given;
+                # rank/left/right above are group-local.  Identity for the
+                # default group, required for subgroups like {1, 2}.
+                left = dist.get_global_rank(group, left)
+                right = dist.get_global_rank(group, right)
             # Post hop 1 before any compute: its wire time hides under the
             # local-block kernel below.  The bidirectional variant drives
             # TWO xGMI point-to-point links per hop (reference
