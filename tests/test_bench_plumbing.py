@@ -76,3 +76,28 @@ def test_example_cpu_smoke():
         capture_output=True, text=True, timeout=300, cwd=REPO)
     assert res.returncode == 0, res.stderr[-2000:]
     assert "done:" in res.stdout
+
+
+def test_bench_distributed_cpu_8proc():
+    """The driver's full-node SCALE launch shape: torch.distributed.run with
+    --nproc-per-node 8 on CPU/gloo — default strategy (bidir ring at W=8:
+    three bidirectional rounds + one remainder hop) and default DDP towers."""
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", "8", "--device", "cpu", "--global-batch", "16", "--dim",
+         "32", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=900, cwd=REPO)
+    assert res.returncode == 0, res.stderr[-2000:]
+    lines = [l for l in res.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, res.stdout
+    out = json.loads(lines[0])
+    assert out["n_gpus"] == 8
+    assert out["config"]["parallelism"] == "dp8-ring"
+    assert out["config"]["ddp_towers"] is True

@@ -94,6 +94,18 @@ def test_w5_bidir_ring_even_hops():
                               rtol=1e-3, atol=1e-6), key
 
 
+def test_w8_bidir_ring_full_node_shape():
+    """W=8 — the 8-GPU node shape the driver's SCALE run uses:
+    divmod(7, 2) = (3, 1), i.e. three bidirectional rounds PLUS a
+    remainder hop (no smaller W hits multiple rounds AND a remainder).
+    Ring raw grads must equal all-gather raw grads."""
+    ring = run_distributed(_ddp_step_raw, 8, "ring", True)[0]
+    gather = run_distributed(_ddp_step_raw, 8, "all_gather", True)[0]
+    for key in ("img", "txt", "scale", "bias", "loss"):
+        assert torch.allclose(ring[key], gather[key],
+                              rtol=1e-3, atol=1e-6), key
+
+
 @pytest.mark.parametrize("diag_offset", [None, 0, 1])
 @pytest.mark.parametrize("col_chunk", [None, 2])
 def test_gradcheck_float64(diag_offset, col_chunk):
