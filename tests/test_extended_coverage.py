@@ -128,6 +128,25 @@ def test_gradcheck_float64(diag_offset, col_chunk):
                                     raise_exception=True)
 
 
+def _sync_comm_step(rank, world):
+    from distributed_sigmoid_loss_amd.parallel.ring import set_sync_comm
+    set_sync_comm(True)
+    try:
+        return _ddp_step_raw(rank, world, "ring", True)
+    finally:
+        set_sync_comm(False)
+
+
+def test_sync_comm_fallback_unchanged():
+    """SIGLIP_SYNC_COMM (the race-bisection debug mode): forcing
+    every exchange to complete synchronously must not change gradients."""
+    sync = run_distributed(_sync_comm_step, 3)[0]
+    normal = run_distributed(_ddp_step_raw, 3, "ring", True)[0]
+    for key in ("img", "txt", "scale", "bias", "loss"):
+        assert torch.allclose(sync[key], normal[key],
+                              rtol=1e-6, atol=1e-8), key
+
+
 def _nograd_step(rank, world, strategy):
     _, _, zi, zt = encode_shard(rank, world, 4, 16)
     mod = DistributedSigmoidLoss(4, strategy=strategy)

@@ -105,7 +105,9 @@ def test_bidir_ring_scaling_oracle(world, batch):
         assert torch.allclose(multi[key], single[key], rtol=1e-3, atol=1e-6), key
 
 
-@pytest.mark.parametrize("world,strategy", [(3, "ring"), (4, "ring_bidir")])
+@pytest.mark.parametrize("world,strategy",
+                         [(3, "ring"), (4, "ring_bidir"),
+                          (8, "ring_bidir")])
 def test_fp8_wire_ring_cpu_routing(world, strategy):
     """fp8 ring ships e4m3 + scale over the wire; on CPU the received
     chunks are dequantized for the torch path.  Validates the wire
