@@ -91,8 +91,11 @@ def _torch_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
     t = t_prime.exp()
     dzimg = torch.zeros_like(zimg)
     dztxt = torch.zeros_like(ztxt)
-    dt_raw = torch.zeros((), dtype=torch.float32, device=zimg.device)
-    dbias = torch.zeros((), dtype=torch.float32, device=zimg.device)
+    # Scalar partials accumulate at >= fp32 (fp64 inputs keep fp64 — the
+    # float64 gradcheck/property oracles compare against autograd exactly).
+    acc = torch.promote_types(zimg.dtype, torch.float32)
+    dt_raw = torch.zeros((), dtype=acc, device=zimg.device)
+    dbias = torch.zeros((), dtype=acc, device=zimg.device)
     step = col_chunk or n
     for j0 in range(0, n, step):
         j1 = min(j0 + step, n)
@@ -104,8 +107,8 @@ def _torch_bwd(zimg, ztxt, t_prime, bias, diag_offset, grad_output,
         g = -lab * torch.sigmoid(-lab * z)
         dzimg += (g @ zt) * t
         dztxt[j0:j1] = (g.T @ zimg) * t
-        dt_raw += (g * dot).float().sum() * t.float()
-        dbias += g.float().sum()
+        dt_raw += (g * dot).to(acc).sum() * t.to(acc)
+        dbias += g.to(acc).sum()
     go = grad_output
     return (dzimg * go, dztxt * go,
             (dt_raw * go).to(t_prime.dtype).reshape(t_prime.shape),
