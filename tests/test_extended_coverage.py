@@ -128,6 +128,33 @@ def test_gradcheck_float64(diag_offset, col_chunk):
                                     raise_exception=True)
 
 
+def _rw_step_w(rank, world, bidir):
+    import torch.nn as nn
+    from distributed_sigmoid_loss_amd import SigLipLoss
+    img_enc, txt_enc, zi, zt = encode_shard(rank, world, 3, 8)
+    scale = nn.Parameter(torch.tensor(math.log(10.0)))
+    bias = nn.Parameter(torch.tensor(-10.0))
+    loss = SigLipLoss(rank=rank, world_size=world, bidir=bidir)(
+        zi, zt, scale, bias)
+    loss.backward()
+    if rank == 0:
+        return {"img": img_enc.weight.grad.clone(),
+                "txt": txt_enc.weight.grad.clone(),
+                "scale": scale.grad.clone(), "bias": bias.grad.clone(),
+                "loss": loss.detach().clone()}
+
+
+@pytest.mark.parametrize("bidir", [True, False])
+def test_sigliploss_chain_w8(bidir):
+    """The reference-API SigLipLoss autograd exchange chain at the full
+    8-rank node shape (7 reversed grad hops unidirectional; 3 bidir rounds
+    + remainder when bidir) vs DistributedSigmoidLoss raw grads."""
+    rw = run_distributed(_rw_step_w, 8, bidir)[0]
+    ddp = run_distributed(_ddp_step_raw, 8, "all_gather", True)[0]
+    for key in ("img", "txt", "scale", "bias", "loss"):
+        assert torch.allclose(rw[key], ddp[key], rtol=1e-3, atol=1e-6), key
+
+
 def _sync_comm_step(rank, world):
     from distributed_sigmoid_loss_amd.parallel.ring import set_sync_comm
     set_sync_comm(True)
