@@ -74,3 +74,38 @@ def test_reduce_out3_layout():
     buf[7, 2] = 8.0
     v = ops.reduce_out3(buf)
     assert v.tolist() == [3.0, 4.0, 8.0]
+
+
+def test_banded_col_step_invariant(monkeypatch):
+    """The banded column step must keep b*step*esz under 32-bit addressing,
+    stay 256-aligned, and respect the SIGLIP_BANDED_STEP override."""
+    monkeypatch.delenv("SIGLIP_BANDED_STEP", raising=False)
+    for b in (131072, 262144, 1048576, 50000):
+        step = ops.banded_col_step(b)
+        assert step % 256 == 0 or step == 256
+        if step > 256:
+            assert b * step * 2 < 2 ** 32
+        # one step wider would break the invariant (maximality)
+        if (step // 256) * 256 == step and b * (step + 256) * 2 < 2 ** 32:
+            assert False, f"step {step} not maximal for b={b}"
+    monkeypatch.setenv("SIGLIP_BANDED_STEP", "512")
+    assert ops.banded_col_step(131072) == 512
+    monkeypatch.setenv("SIGLIP_BANDED_STEP", "7")   # floored to 256
+    assert ops.banded_col_step(131072) == 256
+
+
+def test_save_g_banded_policy_gates(monkeypatch):
+    """Banded saved-g: bf16-only, honors the kill switches, and (without a
+    GPU) reports unusable rather than crashing."""
+    monkeypatch.delenv("SIGLIP_SAVE_G", raising=False)
+    monkeypatch.delenv("SIGLIP_SAVE_G_BANDED", raising=False)
+    assert not ops.save_g_banded_enabled(131072, 131072, "fp8")
+    assert not ops.save_g_banded_enabled(131072, 131072, "mixed")
+    monkeypatch.setenv("SIGLIP_SAVE_G", "0")
+    assert not ops.save_g_banded_enabled(131072, 131072, "bf16")
+    monkeypatch.delenv("SIGLIP_SAVE_G", raising=False)
+    monkeypatch.setenv("SIGLIP_SAVE_G_BANDED", "0")
+    assert not ops.save_g_banded_enabled(131072, 131072, "bf16")
+    monkeypatch.delenv("SIGLIP_SAVE_G_BANDED", raising=False)
+    if not torch.cuda.is_available():
+        assert not ops.save_g_banded_enabled(131072, 131072, "bf16")
